@@ -1,0 +1,245 @@
+// PyTorch bindings for the gfx950 HIP kernels (compiled with hipcc,
+// linked against libtorch; no hipify, no CUDA compatibility layer —
+// c10::hip is the native ROCm stream API).
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void rmsnorm_fwd_launch(const void*, const void*, void*, void*, int64_t, int,
+                        float, hipStream_t);
+void rmsnorm_bwd_launch(const void*, const void*, const void*, const void*,
+                        void*, float*, void*, int, int64_t, int, hipStream_t);
+void rope_launch(const void*, void*, const float*, const float*, const int*,
+                 int64_t, int, int, int, int, hipStream_t);
+void silu_mul_fwd_launch(const void*, void*, int64_t, int, hipStream_t);
+void silu_mul_bwd_launch(const void*, const void*, void*, int64_t, int,
+                         hipStream_t);
+void ce_fwd_launch(const void*, const int64_t*, float*, float*, int64_t, int,
+                   int, int, int64_t, hipStream_t);
+void ce_fwd_sharded_launch(const void*, const int64_t*, float*, float*, float*,
+                           int64_t, int, int, int, int64_t, int64_t,
+                           hipStream_t);
+void ce_bwd_launch(const void*, const int64_t*, const float*, void*, float,
+                   int64_t, int, int, int, int64_t, int64_t, int, hipStream_t);
+void adamw_launch(const void*, int, float, float, float, float, float, float,
+                  float, hipStream_t);
+void attn_fwd_launch(const void*, const void*, const void*, void*, float*,
+                     int, int, int, int, int, float, hipStream_t);
+void attn_bwd_launch(const void*, const void*, const void*, const void*,
+                     const void*, const float*, float*, void*, void*, void*,
+                     int, int, int, int, int, float, hipStream_t);
+}
+
+namespace {
+
+inline hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+#define CHECK_BF16_CONTIG(t)                                          \
+  TORCH_CHECK((t).is_cuda(), #t " must be on the GPU");               \
+  TORCH_CHECK((t).dtype() == torch::kBFloat16, #t " must be bf16");   \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+// ---------------- rmsnorm ----------------
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
+                                       double eps) {
+  CHECK_BF16_CONTIG(x);
+  CHECK_BF16_CONTIG(w);
+  const int H = (int)x.size(-1);
+  const int64_t nrows = x.numel() / H;
+  auto y = torch::empty_like(x);
+  auto rstd = torch::empty({nrows}, x.options().dtype(torch::kFloat));
+  rmsnorm_fwd_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(), rstd.data_ptr(),
+                     nrows, H, (float)eps, cur_stream());
+  return {y, rstd};
+}
+
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor w, torch::Tensor rstd) {
+  CHECK_BF16_CONTIG(dy);
+  CHECK_BF16_CONTIG(x);
+  const int H = (int)x.size(-1);
+  const int64_t nrows = x.numel() / H;
+  const int nblocks = (int)std::min<int64_t>(nrows, 512);
+  auto dx = torch::empty_like(x);
+  auto dw = torch::empty_like(w);
+  auto dw_partial =
+      torch::empty({(int64_t)nblocks, (int64_t)H}, x.options().dtype(torch::kFloat));
+  rmsnorm_bwd_launch(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                     rstd.data_ptr(), dx.data_ptr(),
+                     dw_partial.data_ptr<float>(), dw.data_ptr(), nblocks,
+                     nrows, H, cur_stream());
+  return {dx, dw};
+}
+
+// ---------------- rope ----------------
+torch::Tensor rope(torch::Tensor x, torch::Tensor cos, torch::Tensor sin,
+                   c10::optional<torch::Tensor> positions, bool backward) {
+  CHECK_BF16_CONTIG(x);
+  TORCH_CHECK(x.dim() == 4, "rope expects [B,S,H,D]");
+  const int S = (int)x.size(1), H = (int)x.size(2), D = (int)x.size(3);
+  TORCH_CHECK(cos.is_contiguous() && sin.is_contiguous());
+  TORCH_CHECK(cos.dtype() == torch::kFloat);
+  const int* pos_ptr = nullptr;
+  if (positions.has_value()) {
+    TORCH_CHECK(positions->dtype() == torch::kInt && positions->is_contiguous());
+    TORCH_CHECK(positions->numel() == S, "positions must be [S]");
+    pos_ptr = positions->data_ptr<int>();
+  } else {
+    TORCH_CHECK(cos.size(0) >= S, "rope table shorter than sequence");
+  }
+  auto y = torch::empty_like(x);
+  rope_launch(x.data_ptr(), y.data_ptr(), cos.data_ptr<float>(),
+              sin.data_ptr<float>(), pos_ptr, x.numel() / D, S, H, D,
+              backward ? 1 : 0, cur_stream());
+  return y;
+}
+
+// ---------------- silu_mul ----------------
+torch::Tensor silu_mul_fwd(torch::Tensor gu) {
+  CHECK_BF16_CONTIG(gu);
+  const int I2 = (int)gu.size(-1);
+  TORCH_CHECK(I2 % 16 == 0, "packed gate_up dim must be a multiple of 16");
+  const int I = I2 / 2;
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto y = torch::empty(sizes, gu.options());
+  silu_mul_fwd_launch(gu.data_ptr(), y.data_ptr(), gu.numel() / I2, I,
+                      cur_stream());
+  return y;
+}
+
+torch::Tensor silu_mul_bwd(torch::Tensor dy, torch::Tensor gu) {
+  CHECK_BF16_CONTIG(dy);
+  CHECK_BF16_CONTIG(gu);
+  const int I2 = (int)gu.size(-1);
+  const int I = I2 / 2;
+  auto dgu = torch::empty_like(gu);
+  silu_mul_bwd_launch(dy.data_ptr(), gu.data_ptr(), dgu.data_ptr(),
+                      gu.numel() / I2, I, cur_stream());
+  return dgu;
+}
+
+// ---------------- cross entropy ----------------
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels,
+                                  int64_t S_out, int64_t ignore_index) {
+  CHECK_BF16_CONTIG(logits);
+  TORCH_CHECK(labels.dtype() == torch::kLong && labels.is_contiguous());
+  TORCH_CHECK(logits.dim() == 3, "logits must be [B,S,V]");
+  const int B = (int)logits.size(0), S = (int)logits.size(1),
+            V = (int)logits.size(2);
+  TORCH_CHECK(V % 8 == 0, "vocab must be padded to a multiple of 8");
+  const int64_t nrows = (int64_t)B * S_out;
+  auto loss = torch::empty({nrows}, logits.options().dtype(torch::kFloat));
+  auto lse = torch::empty({nrows}, logits.options().dtype(torch::kFloat));
+  ce_fwd_launch(logits.data_ptr(), labels.data_ptr<int64_t>(),
+                loss.data_ptr<float>(), lse.data_ptr<float>(), nrows, S,
+                (int)S_out, V, ignore_index, cur_stream());
+  return {loss, lse};
+}
+
+std::vector<torch::Tensor> ce_fwd_sharded(torch::Tensor logits,
+                                          torch::Tensor labels, int64_t S_out,
+                                          int64_t vocab_start,
+                                          int64_t ignore_index) {
+  CHECK_BF16_CONTIG(logits);
+  const int B = (int)logits.size(0), S = (int)logits.size(1),
+            V = (int)logits.size(2);
+  const int64_t nrows = (int64_t)B * S_out;
+  auto opts = logits.options().dtype(torch::kFloat);
+  auto maxout = torch::empty({nrows}, opts);
+  auto sumout = torch::empty({nrows}, opts);
+  auto gathered = torch::empty({nrows}, opts);
+  ce_fwd_sharded_launch(logits.data_ptr(), labels.data_ptr<int64_t>(),
+                        maxout.data_ptr<float>(), sumout.data_ptr<float>(),
+                        gathered.data_ptr<float>(), nrows, S, (int)S_out, V,
+                        vocab_start, ignore_index, cur_stream());
+  return {maxout, sumout, gathered};
+}
+
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels,
+                     torch::Tensor lse, double scale, int64_t S_out,
+                     int64_t vocab_start, int64_t ignore_index, bool sharded) {
+  CHECK_BF16_CONTIG(logits);
+  const int B = (int)logits.size(0), S = (int)logits.size(1),
+            V = (int)logits.size(2);
+  const int64_t nrows = (int64_t)B * S_out;
+  auto dlogits = torch::zeros_like(logits);  // non-loss rows stay zero
+  ce_bwd_launch(logits.data_ptr(), labels.data_ptr<int64_t>(),
+                lse.data_ptr<float>(), dlogits.data_ptr(), (float)scale, nrows,
+                S, (int)S_out, V, vocab_start, ignore_index, sharded ? 1 : 0,
+                cur_stream());
+  return dlogits;
+}
+
+// ---------------- adamw ----------------
+void adamw_step(torch::Tensor descs, int64_t nchunks, double lr, double beta1,
+                double beta2, double eps, double wd, int64_t step) {
+  TORCH_CHECK(descs.is_cuda() && descs.dtype() == torch::kLong &&
+              descs.is_contiguous());
+  const float bc1 = 1.0f - std::pow((float)beta1, (float)step);
+  const float bc2 = 1.0f - std::pow((float)beta2, (float)step);
+  adamw_launch(descs.data_ptr(), (int)nchunks, (float)lr, (float)beta1,
+               (float)beta2, (float)eps, (float)wd, 1.0f / bc1, 1.0f / bc2,
+               cur_stream());
+}
+
+// ---------------- attention ----------------
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, double scale) {
+  CHECK_BF16_CONTIG(q);
+  CHECK_BF16_CONTIG(k);
+  CHECK_BF16_CONTIG(v);
+  TORCH_CHECK(q.dim() == 4, "q must be [B,S,Hq,D]");
+  const int B = (int)q.size(0), S = (int)q.size(1), Hq = (int)q.size(2),
+            D = (int)q.size(3);
+  const int Hkv = (int)k.size(2);
+  TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
+  TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({(int64_t)B, (int64_t)Hq, (int64_t)S},
+                          q.options().dtype(torch::kFloat));
+  attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                  lse.data_ptr<float>(), B, S, Hq, Hkv, D, (float)scale,
+                  cur_stream());
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    double scale) {
+  CHECK_BF16_CONTIG(dout);
+  const int B = (int)q.size(0), S = (int)q.size(1), Hq = (int)q.size(2),
+            D = (int)q.size(3);
+  const int Hkv = (int)k.size(2);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  auto delta = torch::empty({(int64_t)B, (int64_t)Hq, (int64_t)S},
+                            q.options().dtype(torch::kFloat));
+  attn_bwd_launch(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                  o.data_ptr(), lse.data_ptr<float>(),
+                  delta.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
+                  dv.data_ptr(), B, S, Hq, Hkv, D, (float)scale,
+                  cur_stream());
+  return {dq, dk, dv};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rope", &rope);
+  m.def("silu_mul_fwd", &silu_mul_fwd);
+  m.def("silu_mul_bwd", &silu_mul_bwd);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_fwd_sharded", &ce_fwd_sharded);
+  m.def("ce_bwd", &ce_bwd);
+  m.def("adamw_step", &adamw_step);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
+}

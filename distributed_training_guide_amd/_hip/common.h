@@ -1,0 +1,91 @@
+// Common device helpers for the MI355X (gfx950 / CDNA4) kernels.
+//
+// Everything here is CDNA4-native: 64-wide wavefronts, bf16 vector loads as
+// short4/short8 reinterprets (hipcc does not auto-vectorize bf16 loads),
+// f32 accumulation, wave shuffle reductions over width 64.
+#pragma once
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <stdint.h>
+
+#define WAVE 64
+
+using bf16 = __hip_bfloat16;
+typedef __attribute__((ext_vector_type(2))) float f32x2;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) short s16x4;
+typedef __attribute__((ext_vector_type(8))) short s16x8;
+
+__device__ __forceinline__ float bf2f(short u) {
+  union { float f; uint32_t i; } c;
+  c.i = ((uint32_t)(uint16_t)u) << 16;
+  return c.f;
+}
+
+__device__ __forceinline__ short f2bf(float f) {
+  // round-to-nearest-even bf16 conversion
+  union { float f; uint32_t i; } c;
+  c.f = f;
+  uint32_t lsb = (c.i >> 16) & 1u;
+  uint32_t rounded = c.i + 0x7fffu + lsb;
+  if ((c.i & 0x7f800000u) == 0x7f800000u) rounded = c.i;  // inf/nan: truncate
+  return (short)(rounded >> 16);
+}
+
+// ---- wave reductions (64 lanes) ----
+__device__ __forceinline__ float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
+  return v;
+}
+
+__device__ __forceinline__ float wave_reduce_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, WAVE));
+  return v;
+}
+
+// ---- block reductions (block size a multiple of 64, <= 1024) ----
+// `scratch` must hold >= blockDim.x/64 floats.
+__device__ __forceinline__ float block_reduce_sum(float v, float* scratch) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int nwaves = blockDim.x / WAVE;
+  v = wave_reduce_sum(v);
+  if (lane == 0) scratch[wid] = v;
+  __syncthreads();
+  float r = (threadIdx.x < nwaves) ? scratch[threadIdx.x] : 0.0f;
+  if (wid == 0) {
+    // reduce across the (<=16) wave partials inside wave 0
+    for (int off = 1; off < nwaves; off <<= 1) r += __shfl_xor(r, off, WAVE);
+    if (lane == 0) scratch[0] = r;
+  }
+  __syncthreads();
+  float out = scratch[0];
+  __syncthreads();
+  return out;
+}
+
+__device__ __forceinline__ float block_reduce_max(float v, float* scratch) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int nwaves = blockDim.x / WAVE;
+  v = wave_reduce_max(v);
+  if (lane == 0) scratch[wid] = v;
+  __syncthreads();
+  float r = (threadIdx.x < nwaves) ? scratch[threadIdx.x] : -INFINITY;
+  if (wid == 0) {
+    for (int off = 1; off < nwaves; off <<= 1) r = fmaxf(r, __shfl_xor(r, off, WAVE));
+    if (lane == 0) scratch[0] = r;
+  }
+  __syncthreads();
+  float out = scratch[0];
+  __syncthreads();
+  return out;
+}
+
+#define HIP_CHECK_DEV(expr)                                                  \
+  do {                                                                       \
+    hipError_t _e = (expr);                                                  \
+    if (_e != hipSuccess) return _e;                                         \
+  } while (0)

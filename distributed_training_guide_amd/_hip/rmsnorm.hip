@@ -1,0 +1,148 @@
+// Fused RMSNorm forward/backward for gfx950.
+//
+// Replaces the reference's LlamaRMSNorm (transformers; invoked from every
+// Llama forward — see SURVEY.md §2b "RMSNorm"). One workgroup per row,
+// bf16x8 vectorized loads (G13), f32 accumulation, wave+block shuffle
+// reductions. Forward saves rstd (f32 per row) for the backward.
+//
+//   y = x * rsqrt(mean(x^2) + eps) * w
+//   dx = rstd * (g - xhat * mean(g * xhat)),  g = dy * w,  xhat = x * rstd
+//   dw = sum_rows dy * xhat   (two-stage: per-block partials, then reduce)
+#include "common.h"
+
+// ---------------- forward ----------------
+__global__ void __launch_bounds__(256) rmsnorm_fwd_kernel(
+    const short* __restrict__ x, const short* __restrict__ w,
+    short* __restrict__ y, float* __restrict__ rstd_out,
+    int64_t nrows, int H, float eps) {
+  __shared__ float scratch[8];
+  const float invH = 1.0f / (float)H;
+  for (int64_t row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const short* xr = x + row * H;
+    short* yr = y + row * H;
+    float ss = 0.0f;
+    if ((H & 7) == 0) {
+      for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+        s16x8 v = *reinterpret_cast<const s16x8*>(xr + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) { float f = bf2f(v[j]); ss += f * f; }
+      }
+    } else {
+      for (int i = threadIdx.x; i < H; i += blockDim.x) {
+        float f = bf2f(xr[i]); ss += f * f;
+      }
+    }
+    ss = block_reduce_sum(ss, scratch);
+    const float rstd = rsqrtf(ss * invH + eps);
+    if (threadIdx.x == 0 && rstd_out != nullptr) rstd_out[row] = rstd;
+    if ((H & 7) == 0) {
+      for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+        s16x8 v = *reinterpret_cast<const s16x8*>(xr + i);
+        s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+        s16x8 o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) o[j] = f2bf(bf2f(v[j]) * rstd * bf2f(wv[j]));
+        *reinterpret_cast<s16x8*>(yr + i) = o;
+      }
+    } else {
+      for (int i = threadIdx.x; i < H; i += blockDim.x)
+        yr[i] = f2bf(bf2f(xr[i]) * rstd * bf2f(w[i]));
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------- backward: dx + per-block dw partials ----------------
+__global__ void __launch_bounds__(256) rmsnorm_bwd_kernel(
+    const short* __restrict__ dy, const short* __restrict__ x,
+    const short* __restrict__ w, const float* __restrict__ rstd,
+    short* __restrict__ dx, float* __restrict__ dw_partial,
+    int64_t nrows, int H) {
+  __shared__ float scratch[8];
+  const float invH = 1.0f / (float)H;
+  float* dwp = dw_partial + (int64_t)blockIdx.x * H;
+  // zero this block's partial row
+  for (int i = threadIdx.x; i < H; i += blockDim.x) dwp[i] = 0.0f;
+  __syncthreads();
+  for (int64_t row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const short* dyr = dy + row * H;
+    const short* xr = x + row * H;
+    short* dxr = dx + row * H;
+    const float rs = rstd[row];
+    // pass 1: dot = mean(g * xhat)
+    float dot = 0.0f;
+    if ((H & 7) == 0) {
+      for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+        s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i);
+        s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i);
+        s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          dot += bf2f(dv[j]) * bf2f(wv[j]) * bf2f(xv[j]) * rs;
+      }
+    } else {
+      for (int i = threadIdx.x; i < H; i += blockDim.x)
+        dot += bf2f(dyr[i]) * bf2f(w[i]) * bf2f(xr[i]) * rs;
+    }
+    dot = block_reduce_sum(dot, scratch) * invH;
+    // pass 2: dx and dw partial accumulate
+    if ((H & 7) == 0) {
+      for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+        s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i);
+        s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i);
+        s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+        s16x8 o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float g = bf2f(dv[j]) * bf2f(wv[j]);
+          float xhat = bf2f(xv[j]) * rs;
+          o[j] = f2bf(rs * (g - xhat * dot));  // dx = rstd*(dy*w - xhat*mean(g*xhat))
+          dwp[i + j] += bf2f(dv[j]) * xhat;
+        }
+        *reinterpret_cast<s16x8*>(dxr + i) = o;
+      }
+    } else {
+      for (int i = threadIdx.x; i < H; i += blockDim.x) {
+        float g = bf2f(dyr[i]) * bf2f(w[i]);
+        float xhat = bf2f(xr[i]) * rs;
+        dxr[i] = f2bf(rs * (g - xhat * dot));
+        dwp[i] += bf2f(dyr[i]) * xhat;
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// reduce dw partials: [nblocks, H] f32 -> dw [H] bf16
+__global__ void __launch_bounds__(256) rmsnorm_dw_reduce_kernel(
+    const float* __restrict__ dw_partial, short* __restrict__ dw,
+    int nblocks, int H) {
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < H;
+       i += gridDim.x * blockDim.x) {
+    float s = 0.0f;
+    for (int b = 0; b < nblocks; ++b) s += dw_partial[(int64_t)b * H + i];
+    dw[i] = f2bf(s);
+  }
+}
+
+extern "C" {
+void rmsnorm_fwd_launch(const void* x, const void* w, void* y, void* rstd,
+                        int64_t nrows, int H, float eps, hipStream_t s) {
+  int grid = (int)(nrows < 2048 ? nrows : 2048);
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(rmsnorm_fwd_kernel, dim3(grid), dim3(256), 0, s,
+                     (const short*)x, (const short*)w, (short*)y, (float*)rstd,
+                     nrows, H, eps);
+}
+void rmsnorm_bwd_launch(const void* dy, const void* x, const void* w,
+                        const void* rstd, void* dx, float* dw_partial,
+                        void* dw, int nblocks, int64_t nrows, int H,
+                        hipStream_t s) {
+  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(nblocks), dim3(256), 0, s,
+                     (const short*)dy, (const short*)x, (const short*)w,
+                     (const float*)rstd, (short*)dx, dw_partial, nrows, H);
+  int rgrid = (H + 255) / 256;
+  hipLaunchKernelGGL(rmsnorm_dw_reduce_kernel, dim3(rgrid), dim3(256), 0, s,
+                     dw_partial, (short*)dw, nblocks, H);
+}
+}

@@ -1,0 +1,18 @@
+"""HIP-kernel-backed ops (GPU) with eager fp32 references (CPU/tests)."""
+from .adamw import FusedAdamW
+from .attention import flash_attention
+from .cross_entropy import causal_lm_loss, sharded_causal_lm_loss
+from .rmsnorm import RMSNorm, rmsnorm
+from .rope import rope
+from .swiglu import silu_mul
+
+__all__ = [
+    "FusedAdamW",
+    "flash_attention",
+    "causal_lm_loss",
+    "sharded_causal_lm_loss",
+    "RMSNorm",
+    "rmsnorm",
+    "rope",
+    "silu_mul",
+]
