@@ -14,7 +14,7 @@ _LLAMA_CONFIGS = {
     "llama-3-70b": (8192, 28672, 80, 64, 8, 128256, 5e5, 8192),
     "llama-3-405b": (16384, 53248, 126, 128, 8, 128256, 5e5, 8192),
     # small configs for tests / smoke runs
-    "llama-debug": (256, 688, 4, 8, 4, 1024, 1e4, 2048),
+    "llama-debug": (256, 688, 4, 4, 2, 1024, 1e4, 2048),
     "llama-60m": (512, 1376, 8, 8, 8, 32000, 1e4, 2048),
 }
 
