@@ -1,0 +1,155 @@
+#!/usr/bin/env python3
+"""Flagship training benchmark — driver contract (BASELINE.json metric).
+
+Measures tokens/sec for a Llama-style causal LM training step (data-parallel
+weak scaling) on N MI355X GPUs: forward + backward + fused-AdamW update on
+synthetic data / random-init weights, bf16 compute, through this repo's HIP
+kernel path and DDP bucket engine over RCCL/xGMI.
+
+    python bench.py --gpus 1 --steps 10 --warmup 3
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
+        --master-addr 127.0.0.1 bench.py --gpus 8 --steps 10 --warmup 3
+
+Rank 0 prints exactly one JSON line: whole-job tokens/s (aggregate over all
+N GPUs), ms_per_step = MAX over ranks, peak HBM, config.  vs_baseline is
+null: the reference's only numeric figure (~137 tok/s/GPU) is its
+Llama-405B / 64xH100 row — a different model/config, not this
+metric's config.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", default="llama-3-8b")
+    p.add_argument("--batch-size", type=int, default=8, help="per GPU")
+    p.add_argument("--seq-length", type=int, default=1024)
+    p.add_argument("--bucket-cap-mb", type=int, default=128)
+    p.add_argument("--no-zero1", dest="zero1", action="store_false",
+                   help="ZeRO-1 optimizer sharding (on by default for N>1)")
+    p.add_argument("--device", default=None)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    import torch.distributed as dist
+
+    from distributed_training_guide_amd.models import build_model, get_config
+    from distributed_training_guide_amd.ops import FusedAdamW
+
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    if args.device:
+        device = torch.device(args.device)
+    elif torch.cuda.is_available():
+        device = torch.device(f"cuda:{local_rank}")
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    if world > 1:
+        from distributed_training_guide_amd.parallel.pg import init_distributed
+
+        init_distributed(device)
+
+    torch.manual_seed(1234 + rank)
+    config = get_config(args.model)
+    model = build_model(config, device=device, dtype=torch.bfloat16)
+    if world > 1:
+        from distributed_training_guide_amd.parallel.ddp import \
+            DistributedDataParallel
+
+        model = DistributedDataParallel(model,
+                                        bucket_cap_mb=args.bucket_cap_mb)
+    if world > 1 and args.zero1:
+        from distributed_training_guide_amd.parallel.zero1 import \
+            ZeroRedundancyOptimizer
+
+        opt = ZeroRedundancyOptimizer(model.parameters(),
+                                      optimizer_class=FusedAdamW, lr=3e-5)
+    else:
+        opt = FusedAdamW(model.parameters(), lr=3e-5)
+
+    B, S = args.batch_size, args.seq_length
+    ids = torch.randint(0, config.vocab_size, (B, S), device=device)
+
+    def step():
+        out = model(input_ids=ids, labels=ids)
+        out.loss.backward()
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+        return out.loss
+
+    def sync():
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+
+    def barrier():
+        if world > 1:
+            dist.barrier()
+
+    for _ in range(args.warmup):
+        step()
+    barrier()
+    sync()
+    t0 = time.time()
+    for _ in range(args.steps):
+        step()
+    barrier()
+    sync()
+    t1 = time.time()
+
+    ms_per_step = (t1 - t0) / args.steps * 1000.0
+    if world > 1:
+        t = torch.tensor([ms_per_step], device=device if device.type == "cuda"
+                         else "cpu", dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        ms_per_step = t.item()
+
+    tokens_per_s = world * B * S / (ms_per_step / 1000.0)
+    peak_gb = (torch.cuda.max_memory_allocated(device) / 1e9
+               if device.type == "cuda" else 0.0)
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "tokens/sec/GPU for Llama-style causal LM at 1/2/4/8 "
+                      "MI355X; peak HBM per GPU",
+            "value": tokens_per_s,
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "peak_hbm_gb": peak_gb,
+            "config": {
+                "model": args.model,
+                "global_batch": B * world,
+                "seq_len": S,
+                "parallelism": f"dp{world}" if world > 1 else "single",
+            },
+        }))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
