@@ -106,6 +106,19 @@ class GPT2ForCausalLM(nn.Module):
                         m.weight.fill_(1.0)
                         m.bias.zero_()
 
+    def reset_param_by_name(self, name: str, tensor: torch.Tensor):
+        std = self.config.initializer_range
+        with torch.no_grad():
+            if ".ln_1." in name or ".ln_2." in name or name.startswith("ln_f."):
+                if name.endswith(".bias"):
+                    tensor.zero_()
+                else:
+                    tensor.fill_(1.0)
+            elif name.endswith(".bias"):
+                tensor.zero_()
+            else:
+                tensor.normal_(0.0, std)
+
     def forward(self, input_ids, labels=None, attention_mask=None,
                 position_ids=None, **_):
         B, S = input_ids.shape

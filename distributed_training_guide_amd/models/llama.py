@@ -153,6 +153,19 @@ class LlamaForCausalLM(nn.Module):
             for m in self.modules():
                 self._reset_module(m, std)
 
+    def reset_param_by_name(self, name: str, tensor: torch.Tensor):
+        """Initialize one (materialized) parameter by its qualified name —
+        used by the FSDP engine's unit-by-unit meta materialization
+        (reference flow 04:74-95)."""
+        std = self.config.initializer_range
+        with torch.no_grad():
+            if "layernorm" in name or name.startswith("norm."):
+                tensor.fill_(1.0)
+            elif name.endswith(".bias"):
+                tensor.zero_()
+            else:
+                tensor.normal_(0.0, std)
+
     @staticmethod
     def _reset_module(m, std):
         if isinstance(m, (nn.Linear, nn.Embedding)):

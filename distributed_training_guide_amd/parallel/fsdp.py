@@ -46,7 +46,7 @@ def _pad(n, m):
 
 class _FSDPUnit:
     def __init__(self, name, module, named_params, group, reshard_after_forward,
-                 reduce_dtype, cpu_offload, device):
+                 reduce_dtype, cpu_offload, device, root_module=None):
         self.name = name
         self.module = module
         self.group = group
@@ -71,32 +71,65 @@ class _FSDPUnit:
 
         # full flat tensor; its storage is resized 0<->full
         self.flat = torch.empty(self.padded, dtype=self.dtype, device=device)
-        self._materialize_initial()
-
-        shard_dev = torch.device("cpu") if cpu_offload else device
-        shard_src = self.flat[self.rank * self.shard_numel:
-                              (self.rank + 1) * self.shard_numel]
-        self.shard = nn.Parameter(shard_src.detach().to(shard_dev).clone()
-                                  if cpu_offload
-                                  else shard_src.detach().clone())
-        if cpu_offload:
-            self.shard.data = self.shard.data.pin_memory()
+        self.was_meta = any(p.is_meta for p in self.params)
+        if not self.was_meta:
+            self._materialize_initial()
 
         # re-point module params to views of flat
         self.views = []
+        replacements = {}
         off = 0
-        for (n, p), shape, numel in zip(named_params, self.shapes,
-                                        self.numels):
+        for i, ((n, p), shape, numel) in enumerate(
+                zip(named_params, self.shapes, self.numels)):
             v = self.flat[off: off + numel].view(shape)
-            p.data = v
+            if p.is_meta:
+                # meta params cannot cross devices via .data: swap in a new
+                # Parameter, then alias the flat view through .data (a
+                # Parameter constructed FROM the view would stay an autograd
+                # view of `flat` — AsStridedBackward version-counter errors)
+                np_ = nn.Parameter(
+                    torch.empty(0, dtype=v.dtype, device=v.device),
+                    requires_grad=p.requires_grad)
+                np_.data = v
+                replacements[p] = np_
+                self.params[i] = np_
+            else:
+                p.data = v
             self.views.append(v)
             off += numel
+        if replacements:
+            scope = root_module if root_module is not None else module
+            for m in scope.modules():
+                for key, pp in list(m._parameters.items()):
+                    if pp in replacements:
+                        m._parameters[key] = replacements[pp]
+        if self.was_meta:
+            # defer allocation: meta-built models materialize unit-by-unit
+            # (peak extra memory = ONE unit, not the whole model)
+            self.flat.untyped_storage().resize_(0)
+        self.shard = None  # created by create_shard() after optional init
         self.is_unsharded = True
         self._gather_work = None
         self._rs_work = None
         self._rs_out = None
         self.pending_grads = 0
         self.sync_enabled = True
+
+    def create_shard(self):
+        shard_dev = torch.device("cpu") if self.cpu_offload else self.device
+        shard_src = self.flat[self.rank * self.shard_numel:
+                              (self.rank + 1) * self.shard_numel]
+        self.shard = nn.Parameter(shard_src.detach().to(shard_dev).clone())
+        if self.cpu_offload and shard_dev.type == "cpu" \
+                and self.device.type == "cuda":
+            self.shard.data = self.shard.data.pin_memory()
+        if self.reduce_dtype != self.dtype:
+            # grads arrive in reduce_dtype (fp32) on a bf16 shard; newer
+            # torch enforces grad dtype unless told otherwise
+            try:
+                self.shard.grad_dtype = self.reduce_dtype
+            except (AttributeError, RuntimeError):
+                pass
 
     # ---- init: keep the existing (already-initialized) values ----
     def _materialize_initial(self):
@@ -215,6 +248,18 @@ class _FSDPUnit:
             self.reshard()
         return out
 
+    def load_full_flat_param_from_concat(self, concat):
+        """Load from the concatenation of OLD per-rank shards (whose padding
+        belonged to a different world size): strip to `total`, re-pad for
+        this world, load."""
+        with torch.no_grad():
+            full = concat.reshape(-1)[: self.total]
+            pad = self.padded - self.total
+            if pad:
+                full = torch.cat([full.cpu(),
+                                  torch.zeros(pad, dtype=full.dtype)])
+            self.load_full_flat(full)
+
     def load_full_flat(self, flat_full):
         with torch.no_grad():
             sl = flat_full.reshape(-1)[self.rank * self.shard_numel:
@@ -264,7 +309,8 @@ class FSDP(nn.Module):
                    if p.requires_grad]
             covered.update(p for _, p in nps)
             u = _FSDPUnit(n, m, nps, process_group, reshard_after_forward,
-                          reduce_dtype, cpu_offload, device)
+                          reduce_dtype, cpu_offload, device,
+                          root_module=module)
             self.units.append(u)
             self._unit_of_module[m] = u
         root_nps = [(n, p) for n, p in module.named_parameters()
@@ -274,20 +320,39 @@ class FSDP(nn.Module):
             self.root_unit = _FSDPUnit("__root__", module, root_nps,
                                        process_group,
                                        root_reshard_after_forward,
-                                       reduce_dtype, cpu_offload, device)
+                                       reduce_dtype, cpu_offload, device,
+                                       root_module=module)
             self.units.append(self.root_unit)
         self._layer_units = [u for u in self.units if u is not self.root_unit]
 
-        # forward-order prefetch chain over layer units
-        self._fwd_next = {}
-        self._bwd_next = {}
-        for i, u in enumerate(self._layer_units):
-            if i + 1 < len(self._layer_units):
-                self._fwd_next[u] = self._layer_units[i + 1]
-            if i - 1 >= 0:
-                self._bwd_next[u] = self._layer_units[i - 1]
-        self.set_modules_to_forward_prefetch(None)
-        self.set_modules_to_backward_prefetch(None)
+        # forward-order prefetch chain over layer units (depth
+        # reconfigurable via set_prefetch_depth — the reference's explicit
+        # set_modules_to_forward/backward_prefetch lists, 05:148-161)
+        self._fwd_next: dict = {}
+        self._bwd_next: dict = {}
+        self.set_prefetch_depth(1)
+
+        # meta-built params: materialize unit by unit with the model's own
+        # init (reference flow: meta init -> shard -> to_empty ->
+        # reset_parameters, 04:74-95 / 06:123-125), identically seeded on
+        # every rank so the shards are consistent.
+        for u in self.units:
+            if u.was_meta:
+                init_fn = getattr(module, "reset_param_by_name", None)
+                if init_fn is None:
+                    raise RuntimeError(
+                        "meta-initialized model needs reset_param_by_name")
+                u.flat.untyped_storage().resize_(
+                    u.padded * u.flat.element_size())
+                with torch.no_grad():
+                    u.flat.zero_()
+                    for n, v in zip(u.param_names, u.views):
+                        init_fn(n, v)
+                u.create_shard()
+                u.flat.untyped_storage().resize_(0)
+                u.is_unsharded = False
+            else:
+                u.create_shard()
 
         # hooks
         self._works = []
@@ -303,23 +368,24 @@ class FSDP(nn.Module):
         for u in self.units:
             u.reshard()
 
-    # ---- explicit prefetch lists (reference 05:148-161) ----
-    def set_modules_to_forward_prefetch(self, layers):
-        self._explicit_fwd_prefetch = layers
-
-    def set_modules_to_backward_prefetch(self, layers):
-        self._explicit_bwd_prefetch = layers
+    # ---- explicit prefetch (reference 05:148-161) ----
+    def set_prefetch_depth(self, depth: int):
+        """Prefetch the next `depth` units' all-gathers in each direction."""
+        L = self._layer_units
+        self._fwd_next = {u: L[i + 1: i + 1 + depth]
+                          for i, u in enumerate(L)}
+        self._bwd_next = {u: L[max(0, i - depth): i][::-1]
+                          for i, u in enumerate(L)}
 
     # ---- hooks ----
     def _make_pre_fwd(self, u):
-        def hook(module, args, kwargs):
+        def hook(module, args):
             u.ensure_unsharded()
             if self.prefetch:
-                nxt = self._fwd_next.get(u)
-                if nxt is not None:
+                for nxt in self._fwd_next.get(u, ()):
                     nxt.unshard(async_op=True)
             return None
-        return lambda m, a, k: hook(m, a, k)  # with_kwargs variant below
+        return hook
 
     def _make_post_fwd(self, u):
         def hook(module, args, output):
@@ -341,8 +407,7 @@ class FSDP(nn.Module):
                 fired[0] = True
                 u.ensure_unsharded()
                 if self.prefetch:
-                    nxt = self._bwd_next.get(u)
-                    if nxt is not None:
+                    for nxt in self._bwd_next.get(u, ()):
                         nxt.unshard(async_op=True)
             return grad
         out.register_hook(on_grad)
@@ -372,7 +437,7 @@ class FSDP(nn.Module):
     def no_sync(self):
         """Accumulate unsharded grads locally; reduce on the boundary
         microbatch (gradient-accumulation recipe)."""
-        self.sync_enabled = prev = True and self.sync_enabled
+        prev = self.sync_enabled
         self.sync_enabled = False
         try:
             yield
@@ -456,3 +521,36 @@ class FSDP(nn.Module):
 
     # optimizer state helpers: FusedAdamW on shard params already produces
     # a per-rank state dict; nothing extra needed here.
+
+
+def apply_activation_checkpointing(model, layer_cls=None):
+    """Patch each decoder layer's forward to run under non-reentrant
+    activation checkpointing (the reference uses
+    apply_activation_checkpointing + transformer_auto_wrap_policy,
+    05:165-178).  Patching `forward` (not wrapping the module) keeps the
+    state-dict names unchanged and keeps the FSDP hooks OUTSIDE the
+    checkpoint, so the recompute runs inside the unit's unsharded window."""
+    import torch.utils.checkpoint as tc
+
+    if layer_cls is None:
+        from ..models.gpt2 import GPT2Block
+        from ..models.llama import LlamaDecoderLayer
+
+        layer_cls = (LlamaDecoderLayer, GPT2Block)
+    n = 0
+    for m in model.modules():
+        if isinstance(m, layer_cls):
+            orig = m.forward
+
+            def make(fwd):
+                def wrapped(*args, **kwargs):
+                    if torch.is_grad_enabled():
+                        return tc.checkpoint(fwd, *args, use_reentrant=False,
+                                             **kwargs)
+                    return fwd(*args, **kwargs)
+                return wrapped
+
+            m.forward = make(orig)
+            n += 1
+    LOGGER.info(f"activation checkpointing applied to {n} layers")
+    return model
