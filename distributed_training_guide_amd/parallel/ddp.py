@@ -103,7 +103,7 @@ class DistributedDataParallel(nn.Module):
         with torch.no_grad():
             for t in list(self.module.parameters()) + \
                     list(self.module.buffers()):
-                dist.broadcast(t.data, src=0, group=self.group)
+                dist.broadcast(t.data, group_src=0, group=self.group)
 
     def _install_grad_views(self):
         for b in self.buckets:

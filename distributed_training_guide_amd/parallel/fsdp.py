@@ -515,7 +515,7 @@ class FSDP(nn.Module):
                 if off < u.padded:
                     flat[off:].zero_()
             if broadcast_from_rank0 and self.world > 1:
-                dist.broadcast(flat, src=0, group=self.group)
+                dist.broadcast(flat, group_src=0, group=self.group)
             u.load_full_flat(flat)
             del flat
 
