@@ -19,6 +19,8 @@ struct AdamWDesc {
   int64_t g_is_bf16;
 };
 
+typedef __attribute__((ext_vector_type(4))) short s16x4v;
+
 __global__ void __launch_bounds__(256) adamw_kernel(
     const AdamWDesc* __restrict__ descs, int nchunks, float lr, float beta1,
     float beta2, float eps, float wd, float inv_bc1, float inv_bc2) {
@@ -29,14 +31,53 @@ __global__ void __launch_bounds__(256) adamw_kernel(
     const bool pbf = d.p_is_bf16 != 0;
     const bool gbf = d.g_is_bf16 != 0;
     const float decay = 1.0f - lr * wd;
-    for (int64_t i = threadIdx.x; i < d.n; i += blockDim.x) {
+    // vectorized main body: 4 elements per thread per iteration (16 B f32
+    // moments, 8/16 B params+grads — memory-bound, G13 vectorization)
+    const int64_t n4 = d.n / 4;
+    for (int64_t q = threadIdx.x; q < n4; q += blockDim.x) {
+      int64_t i = q * 4;
+      f32x4 g4, p4;
+      if (gbf) {
+        s16x4v gv = *reinterpret_cast<const s16x4v*>((const short*)d.g + i);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) g4[j] = bf2f(gv[j]);
+      } else {
+        g4 = *reinterpret_cast<const f32x4*>((const float*)d.g + i);
+      }
+      if (pbf) {
+        s16x4v pv = *reinterpret_cast<const s16x4v*>((const short*)d.p + i);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) p4[j] = bf2f(pv[j]);
+      } else {
+        p4 = *reinterpret_cast<const f32x4*>((const float*)d.p + i);
+      }
+      f32x4 m4 = *reinterpret_cast<const f32x4*>(m + i);
+      f32x4 v4 = *reinterpret_cast<const f32x4*>(v + i);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        m4[j] = beta1 * m4[j] + (1.0f - beta1) * g4[j];
+        v4[j] = beta2 * v4[j] + (1.0f - beta2) * g4[j] * g4[j];
+        p4[j] = p4[j] * decay -
+                lr * (m4[j] * inv_bc1) / (sqrtf(v4[j] * inv_bc2) + eps);
+      }
+      *reinterpret_cast<f32x4*>(m + i) = m4;
+      *reinterpret_cast<f32x4*>(v + i) = v4;
+      if (pbf) {
+        s16x4v pv;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) pv[j] = f2bf(p4[j]);
+        *reinterpret_cast<s16x4v*>((short*)d.p + i) = pv;
+      } else {
+        *reinterpret_cast<f32x4*>((float*)d.p + i) = p4;
+      }
+    }
+    // scalar tail
+    for (int64_t i = n4 * 4 + threadIdx.x; i < d.n; i += blockDim.x) {
       float g = gbf ? bf2f(((const short*)d.g)[i]) : ((const float*)d.g)[i];
       float p = pbf ? bf2f(((const short*)d.p)[i]) : ((const float*)d.p)[i];
       float mi = m[i] = beta1 * m[i] + (1.0f - beta1) * g;
       float vi = v[i] = beta2 * v[i] + (1.0f - beta2) * g * g;
-      float mhat = mi * inv_bc1;
-      float vhat = vi * inv_bc2;
-      p = p * decay - lr * mhat / (sqrtf(vhat) + eps);
+      p = p * decay - lr * (mi * inv_bc1) / (sqrtf(vi * inv_bc2) + eps);
       if (pbf) ((short*)d.p)[i] = f2bf(p);
       else ((float*)d.p)[i] = p;
     }

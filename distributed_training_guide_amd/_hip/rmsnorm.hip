@@ -119,9 +119,17 @@ __global__ void __launch_bounds__(256) rmsnorm_dw_reduce_kernel(
     int nblocks, int H) {
   for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < H;
        i += gridDim.x * blockDim.x) {
-    float s = 0.0f;
-    for (int b = 0; b < nblocks; ++b) s += dw_partial[(int64_t)b * H + i];
-    dw[i] = f2bf(s);
+    // 8 independent accumulator chains: the row loop is latency-bound
+    // otherwise (one dependent L2/HBM load per iteration)
+    float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    int b = 0;
+    for (; b + 8 <= nblocks; b += 8)
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        s[j] += dw_partial[(int64_t)(b + j) * H + i];
+    for (; b < nblocks; ++b) s[0] += dw_partial[(int64_t)b * H + i];
+    dw[i] = f2bf(((s[0] + s[1]) + (s[2] + s[3])) +
+                 ((s[4] + s[5]) + (s[6] + s[7])));
   }
 }
 

@@ -1,0 +1,52 @@
+#!/usr/bin/env python3
+"""Elastic-restart demonstration — counterpart of
+/root/reference/related-topics/elastic-training/toy.py: each rank randomly
+crashes; torchrun --max-restarts restarts ALL workers; progress survives in
+a shared state file written by rank 0 under barrier fencing.  No GPU
+required (gloo).
+
+    torchrun --standalone --local-addr 127.0.0.1 --nproc-per-node 4 \
+        --max-restarts 3 related-topics/elastic-training/toy.py
+"""
+import json
+import os
+import random
+import time
+
+import torch.distributed as dist
+from torch.distributed.elastic.multiprocessing.errors import record
+
+STATE_FILE = os.environ.get("TOY_STATE_FILE", "toy-state.json")
+
+
+@record
+def main():
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    dist.init_process_group("gloo")
+    # NOTE: rank/world_size are NOT stable across restarts — reload shared
+    # progress from the state file, never from process memory.
+    state = {"iteration": 0}
+    if os.path.exists(STATE_FILE):
+        with open(STATE_FILE) as fp:
+            state = json.load(fp)
+    print(f"[rank={rank}/{world} restart={os.environ.get('TORCHELASTIC_RESTART_COUNT', 0)}] "
+          f"resuming at iteration {state['iteration']}")
+
+    for it in range(state["iteration"], 20):
+        time.sleep(0.1)  # "training"
+        if random.random() < 0.05:
+            raise RuntimeError(f"rank {rank} simulated failure at iter {it}")
+        state["iteration"] = it + 1
+        dist.barrier()
+        if rank == 0:
+            with open(STATE_FILE, "w") as fp:
+                json.dump(state, fp)
+        dist.barrier()
+    if rank == 0:
+        print("done:", state)
+        os.unlink(STATE_FILE)
+
+
+if __name__ == "__main__":
+    main()
