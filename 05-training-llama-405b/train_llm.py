@@ -33,6 +33,9 @@ def build_parser():
     p.add_argument("--broadcast-init", action="store_true",
                    help="init via rank-0 full state dict broadcast "
                         "(the pretrained-weights path of the reference)")
+    p.add_argument("--weights-dir", default=None,
+                   help="node-local safetensors dir written by download.py; "
+                        "rank 0 loads on CPU and broadcasts shards")
     return p
 
 

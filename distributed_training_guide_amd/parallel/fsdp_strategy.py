@@ -47,7 +47,17 @@ class FSDPStrategy:
                      reduce_dtype=torch.float32)
         depth = getattr(args, "prefetch_layers", 1) or 1
         model.set_prefetch_depth(depth)
-        if getattr(args, "broadcast_init", False):
+        weights_dir = getattr(args, "weights_dir", None)
+        if weights_dir:
+            # ch-5 pretrained path: rank 0 loads node-local safetensors
+            # shards (written by 05-.../download.py) on CPU and broadcasts
+            # (reference 05:76-85 from_pretrained + 05:118-126 broadcast)
+            full = {}
+            if self.rank == 0:
+                full = _load_safetensors_dir(Path(weights_dir),
+                                             args.model_name)
+            model.load_full_state_dict(full, broadcast_from_rank0=True)
+        elif getattr(args, "broadcast_init", False):
             # exercise the ch-5 pretrained-init path: rank 0 materializes a
             # full state dict on CPU and broadcasts shards (05:118-126)
             full = model.full_state_dict(rank0_only=True, offload_to_cpu=True)
@@ -138,3 +148,20 @@ def _load_resharding(exp_dir: Path, model, optimizer, rank):
     optimizer.load_state_dict(new_sd)
     with open(exp_dir / "state.json") as fp:
         return json.load(fp)
+
+
+def _load_safetensors_dir(weights_dir: Path, model_name: str):
+    """Load a node-local safetensors shard directory written by
+    05-training-llama-405b/download.py into a full CPU state dict
+    (reference 05:76-85: from_pretrained on rank 0 CPU only)."""
+    import json as _json
+
+    from safetensors.torch import load_file
+
+    d = weights_dir / model_name if (weights_dir / model_name).is_dir() \
+        else weights_dir
+    index = _json.loads((d / "model.safetensors.index.json").read_text())
+    full = {}
+    for fname in sorted(set(index["weight_map"].values())):
+        full.update(load_file(str(d / fname)))
+    return full

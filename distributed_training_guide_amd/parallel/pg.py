@@ -35,6 +35,10 @@ def init_distributed(device: torch.device | None = None,
     rank, local_rank, world = env_rank(), env_local_rank(), env_world_size()
     if dist.is_initialized():
         return dist.get_rank(), local_rank, dist.get_world_size()
+    if world == 1 and "MASTER_ADDR" not in os.environ:
+        # single-process launch without torchrun: self-rendezvous
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29511")
     backend = "nccl" if (device is not None and device.type == "cuda") else \
         ("nccl" if torch.cuda.is_available() else "gloo")
     kwargs = dict(rank=rank, world_size=world,
