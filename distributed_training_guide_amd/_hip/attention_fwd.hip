@@ -1,5 +1,5 @@
-// Causal flash-attention forward for gfx950 (MFMA 16x16x32 bf16, online
-// softmax) — v2: LDS-staged K/V tiles with XOR swizzle.
+// Causal flash-attention forward for gfx950 — v3: swapped-operand S^T MFMA
+// with in-register P, direct-from-L2 K, transposed double-buffered V LDS.
 //
 // Replaces the reference's flash-attn-2 dependency
 // (attn_implementation="flash_attention_2", 05:93 / 06:73 / 07:71 —
@@ -8,16 +8,26 @@
 // Layout: q [B,S,Hq,D], k/v [B,S,Hkv,D] bf16 contiguous (BSHD). GQA via
 // Hq % Hkv == 0. Causal always. Saves lse [B,Hq,S] f32 for the backward.
 //
-// Structure:
-//   grid = B * Hq * ceil(S/128); block = 256 threads = 4 waves.
-//   Each wave owns 32 query rows (Q fragments live in registers for the
-//   whole kernel); KV tiles of 64 keys are cooperatively staged into LDS
-//   (vectorized 16 B global loads, byte ^= (row&7)<<4 slot swizzle so the
-//   per-lane-row ds_read_b128 fragment reads are conflict-free — guide §6
-//   G4) and shared by all 4 waves. Per tile: S = scale*QK^T (32 MFMA),
-//   online softmax in the C-fragment layout, P staged through a per-wave
-//   swizzled LDS tile to re-enter MFMA as the A operand, O += P.V
-//   (32 MFMA) with the standard rescale.
+// v3 design (guide §5.5 / §5 "common mistakes"):
+//   * S^T = mfma(A=K, B=Q): the C fragment then holds qrow = lane&15 —
+//     softmax rows are (mostly) lane-local: 16 values in registers + two
+//     shfl_xor steps, instead of 16-lane tree reductions per row.
+//   * Key-permuted A fragments: K rows are fed to the QK^T MFMA in an
+//     order chosen so the C-fragment's per-lane key positions equal the
+//     B-operand layout that P^T needs for the P·V MFMA. P therefore never
+//     leaves registers: exp -> v_cvt_pk_bf16_f32 -> B fragment. No P LDS
+//     round-trip, no block barrier between QK^T and P·V.
+//   * K is NOT staged in LDS: at training shapes the K tile is L1/L2
+//     resident (guide §5 mistake 7) and the A-fragment read of K is a
+//     clean bf16x8 global load (8 consecutive d per lane). The XCD-aware
+//     block remap below keeps all blocks of one (b, kv-head) on one XCD's
+//     L2.
+//   * V is staged TRANSPOSED into double-buffered LDS (v^T[d][key], key
+//     index XOR-swizzled): the P·V MFMA's A operand needs 8 consecutive
+//     keys per lane, which only a transposed image can serve as
+//     ds_read_b128. One barrier per KV tile (ping-pong buffers).
+//   * O^T accumulates in C fragments (qrow = lane&15 throughout), scale
+//     folded into exp2f, epilogue writes 4 consecutive d per lane (b64).
 //
 // MFMA fragment maps (gfx950 v_mfma_f32_16x16x32_bf16):
 //   A[m][k]: lane l holds m = l&15, k = (l>>4)*8 + j (j=0..7)
@@ -26,248 +36,268 @@
 #include "common.h"
 
 using bf16x8 = s16x8;
+typedef __attribute__((ext_vector_type(2))) short s16x2;
+typedef __attribute__((ext_vector_type(4))) short s16x4v;
 
 __device__ __forceinline__ f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
-__device__ __forceinline__ float group16_max(float v) {
-#pragma unroll
-  for (int off = 1; off < 16; off <<= 1) v = fmaxf(v, __shfl_xor(v, off, WAVE));
-  return v;
-}
-__device__ __forceinline__ float group16_sum(float v) {
-#pragma unroll
-  for (int off = 1; off < 16; off <<= 1) v += __shfl_xor(v, off, WAVE);
-  return v;
+// pack two f32 into one reg of 2 bf16 (no builtin on gfx950 — guide T12)
+__device__ __forceinline__ uint32_t cvt_pk_bf16(float lo, float hi) {
+  uint32_t r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
 }
 
-// swizzled element index into a [rows][D] bf16 LDS tile (16B slot XOR)
-__device__ __forceinline__ int swz(int row, int d, int D) {
-  return row * D + (((d >> 3) ^ (row & 7)) << 3) + (d & 7);
+#define QBLK 128   // q rows per block (32 per wave)
+#define KVBLK 64   // keys per LDS tile
+#define LOG2E 1.4426950408889634f
+
+// v^T LDS image: [D up to 128][KVBLK] bf16, key index swizzled by d so the
+// transposed staging writes spread banks while the per-d 8-key block read
+// stays 16B-contiguous.
+__device__ __forceinline__ int vt_idx(int d, int key) {
+  return d * KVBLK + (key ^ ((d & 7) << 3) ^ (((d >> 3) & 7) << 3));
 }
 
-#define QBLK 128  // q rows per block (32 per wave)
-#define KVBLK 64  // keys per LDS tile
-
-__global__ void __launch_bounds__(256) attn_fwd_kernel(
+__global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
     float* __restrict__ lse_out, int B, int S, int Hq, int Hkv, int D,
     float scale) {
-  __shared__ short k_lds[KVBLK * 128];
-  __shared__ short v_lds[KVBLK * 128];
-  __shared__ short p_lds[4][32 * KVBLK];  // per-wave [32 q][64 k], swizzled
+  __shared__ short vt_lds[2][128 * KVBLK];
 
-  const int ntiles_q = (S + QBLK - 1) / QBLK;
-  const int bid = blockIdx.x;
-  const int qtile = bid % ntiles_q;
-  const int h = (bid / ntiles_q) % Hq;
-  const int b = bid / (ntiles_q * Hq);
-  const int hkv = h / (Hq / Hkv);
+  const int ntq = (S + QBLK - 1) / QBLK;
+  const int gqa = Hq / Hkv;
+  const int bpg = gqa * ntq;  // blocks per (b, hkv) group
+
+  // XCD-aware remap (guide T1): all blocks of one (b,hkv) group land on one
+  // XCD so its K/V stay in that XCD's L2. Dispatcher places block i on XCD
+  // i%8 (performance only, never correctness).
+  int b, h, qtile;
+  {
+    const int NG = B * Hkv;
+    int bid = blockIdx.x;
+    int gid, within;
+    if ((NG & 7) == 0) {
+      gid = (bid >> 3) / bpg * 8 + (bid & 7);
+      within = (bid >> 3) % bpg;
+    } else {
+      gid = bid / bpg;
+      within = bid % bpg;
+    }
+    b = gid / Hkv;
+    const int hkv = gid % Hkv;
+    h = hkv * gqa + within / ntq;
+    qtile = within % ntq;
+  }
+  const int hkv = h / gqa;
 
   const int tid = threadIdx.x;
-  const int wid = tid / WAVE;
+  const int wid = tid >> 6;
   const int lane = tid & (WAVE - 1);
   const int l15 = lane & 15;
   const int lg = lane >> 4;
 
   const int q0 = qtile * QBLK + wid * 32;  // this wave's 32 q rows
-  const int nd = D / 16, ndk = D / 32;
+  const int nd16 = D >> 4;                 // 16-d tiles (8 for D=128)
+  const int nkc = D >> 5;                  // 32-d MFMA K chunks
   const int64_t strideS_q = (int64_t)Hq * D;
   const int64_t strideS_kv = (int64_t)Hkv * D;
   const short* qb = q + ((int64_t)b * S * Hq + h) * D;
   const short* kb = k + ((int64_t)b * S * Hkv + hkv) * D;
   const short* vb = v + ((int64_t)b * S * Hkv + hkv) * D;
 
-  // ---- Q fragments: [2 mh][ndk chunks], rows q0 + mh*16 + l15 ----
+  // ---- Q B-fragments: [nq 2][kc], row q0 + nq*16 + l15, d = kc*32+lg*8 --
   bf16x8 qf[2][4];
 #pragma unroll
-  for (int mh = 0; mh < 2; ++mh) {
-    int qrow = q0 + mh * 16 + l15;
+  for (int nq = 0; nq < 2; ++nq) {
+    int qrow = q0 + nq * 16 + l15;
     if (qrow >= S) qrow = S - 1;
-    const short* qp = qb + (int64_t)qrow * strideS_q;
+    const short* qp = qb + (int64_t)qrow * strideS_q + lg * 8;
 #pragma unroll
-    for (int c = 0; c < 4; ++c)
-      if (c < ndk)
-        qf[mh][c] = *reinterpret_cast<const bf16x8*>(qp + c * 32 + lg * 8);
+    for (int kc = 0; kc < 4; ++kc)
+      if (kc < nkc)
+        qf[nq][kc] = *reinterpret_cast<const bf16x8*>(qp + kc * 32);
   }
 
-  f32x4 oacc[2][8];
+  // key permutation: C-fragment m-position p (= mt*16 + lg*4 + r) must hold
+  // key (mt>>1)*32 + lg*8 + (mt&1)*4 + r, so the packed P rows are already
+  // the B-operand key order (kc*32 + lg*8 + j). A-operand lane position is
+  // l15 = lg_c*4 + r of the C tile.
+  int kperm[4];  // key offset of A-fragment row l15 in tile mt
 #pragma unroll
-  for (int mh = 0; mh < 2; ++mh)
+  for (int mt = 0; mt < 4; ++mt)
+    kperm[mt] = (mt >> 1) * 32 + (l15 >> 2) * 8 + (mt & 1) * 4 + (l15 & 3);
+  // per-lane key offsets of the C fragment (for causal masking):
+  // key(mt, r) = (mt>>1)*32 + lg*8 + (mt&1)*4 + r
+  int ckey[4];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) oacc[mh][i] = {0.f, 0.f, 0.f, 0.f};
-  float mrow[2][4], lrow[2][4];
+  for (int mt = 0; mt < 4; ++mt)
+    ckey[mt] = (mt >> 1) * 32 + lg * 8 + (mt & 1) * 4;
+
+  // O^T accumulators: [d tile][nq] C frags; lane: qrow=l15, d=lg*4+r
+  f32x4 oacc[8][2];
 #pragma unroll
-  for (int mh = 0; mh < 2; ++mh)
+  for (int dt = 0; dt < 8; ++dt)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      mrow[mh][r] = -INFINITY;
-      lrow[mh][r] = 0.f;
+    for (int nq = 0; nq < 2; ++nq) oacc[dt][nq] = {0.f, 0.f, 0.f, 0.f};
+  float mrow[2] = {-INFINITY, -INFINITY};
+  float lrow[2] = {0.f, 0.f};
+
+  const int kv_end = min(S, qtile * QBLK + QBLK);
+  const int ntiles = (kv_end + KVBLK - 1) / KVBLK;
+  const float c = scale * LOG2E;
+
+  // ---- staging: thread stages V rows (key = tid&63), 16B of d each ----
+  const int skey = tid & 63;
+  const int sslot0 = (tid >> 6) * 2;  // wave w covers d slots {2w,2w+1,8+2w,8+2w+1}
+  auto stage_v = [&](int kv0, int buf) {
+    int keyg = kv0 + skey;
+    if (keyg >= S) keyg = S - 1;
+    const short* vp = vb + (int64_t)keyg * strideS_kv;
+    short* dst = vt_lds[buf];
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+#pragma unroll
+      for (int ss = 0; ss < 2; ++ss) {
+        int slot = half * 8 + sslot0 + ss;  // d0 = slot*8
+        if (slot * 8 >= D) break;
+        bf16x8 vec = *reinterpret_cast<const bf16x8*>(vp + slot * 8);
+        int d0 = slot * 8;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) dst[vt_idx(d0 + i, skey)] = vec[i];
+      }
     }
+  };
 
-  const int kv_end = min(S, qtile * QBLK + QBLK);  // block-uniform causal
-  const int nvec = KVBLK * D / 8 / 256;            // 16B vectors per thread
+  stage_v(0, 0);
+  __syncthreads();
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
-    // ---- cooperative staging: K and V tiles, swizzled ----
-#pragma unroll
-    for (int vv = 0; vv < 4; ++vv) {
-      if (vv >= nvec) break;
-      int vecid = vv * 256 + tid;
-      int key = vecid / (D / 8);
-      int slot = vecid % (D / 8);
-      int keyg = kv0 + key;
-      if (keyg >= S) keyg = S - 1;
-      const short* kp = kb + (int64_t)keyg * strideS_kv + slot * 8;
-      const short* vp = vb + (int64_t)keyg * strideS_kv + slot * 8;
-      int dst = key * D + (((slot ^ (key & 7)) & (D / 8 - 1)) << 3);
-      *reinterpret_cast<bf16x8*>(k_lds + dst) =
-          *reinterpret_cast<const bf16x8*>(kp);
-      *reinterpret_cast<bf16x8*>(v_lds + dst) =
-          *reinterpret_cast<const bf16x8*>(vp);
-    }
-    __builtin_amdgcn_s_barrier();
+  for (int t = 0; t < ntiles; ++t) {
+    const int kv0 = t * KVBLK;
+    const int buf = t & 1;
+    // stage next tile into the other buffer (global loads issue before the
+    // MFMA stream — T14 spirit; ds_writes don't touch the compute buffer)
+    if (t + 1 < ntiles) stage_v((t + 1) * KVBLK, buf ^ 1);
 
-    // waves fully left of the diagonal skip compute (barriers stay uniform)
     const bool active = (kv0 <= q0 + 31);
-
-    // ---- S = scale * Q K^T : [2 mh][4 ntiles] C-frags ----
-    f32x4 sfrag[2][4];
-    float alpha[2][4];
     if (active) {
+      // ---- S^T = mfma(K, Q): [mt 4][nq 2] C frags ----
+      f32x4 sfrag[4][2];
 #pragma unroll
-    for (int mh = 0; mh < 2; ++mh)
+      for (int mt = 0; mt < 4; ++mt) {
+        int keyg = kv0 + kperm[mt];
+        if (keyg >= S) keyg = S - 1;
+        const short* kp = kb + (int64_t)keyg * strideS_kv + lg * 8;
+        f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
-        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-        int key = nt * 16 + l15;
-#pragma unroll
-        for (int c = 0; c < 4; ++c)
-          if (c < ndk) {
-            int d = c * 32 + lg * 8;
-            bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-                k_lds + swz(key, d, D));
-            acc = mfma16(qf[mh][c], kf, acc);
+        for (int kc = 0; kc < 4; ++kc)
+          if (kc < nkc) {
+            bf16x8 kf = *reinterpret_cast<const bf16x8*>(kp + kc * 32);
+            acc0 = mfma16(kf, qf[0][kc], acc0);
+            acc1 = mfma16(kf, qf[1][kc], acc1);
           }
-        sfrag[mh][nt] = acc;
+        sfrag[mt][0] = acc0;
+        sfrag[mt][1] = acc1;
       }
 
-    // ---- causal mask + online softmax (C layout) ----
+      // ---- causal mask + online softmax (rows lane-local) ----
+      const bool need_mask = (kv0 + KVBLK - 1) > q0 || kv_end < kv0 + KVBLK;
+      uint32_t pk[2][2][4];  // [nq][kc][4 regs of 2 bf16]
+      float alpha[2];
 #pragma unroll
-    for (int mh = 0; mh < 2; ++mh) {
-      float tmax[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+      for (int nq = 0; nq < 2; ++nq) {
+        const int qrow = q0 + nq * 16 + l15;
+        if (need_mask) {
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
-        int key = kv0 + nt * 16 + l15;
+          for (int mt = 0; mt < 4; ++mt)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int qrow = q0 + mh * 16 + lg * 4 + r;
-          float s = sfrag[mh][nt][r] * scale;
-          if (key > qrow || key >= S) s = -INFINITY;
-          sfrag[mh][nt][r] = s;
-          tmax[r] = fmaxf(tmax[r], s);
+            for (int r = 0; r < 4; ++r) {
+              int key = kv0 + ckey[mt] + r;
+              if (key > qrow || key >= S) sfrag[mt][nq][r] = -INFINITY;
+            }
         }
-      }
+        float tmax = -INFINITY;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        tmax[r] = group16_max(tmax[r]);
-        float mnew = fmaxf(mrow[mh][r], tmax[r]);
-        alpha[mh][r] =
-            (mrow[mh][r] == -INFINITY) ? 0.0f : __expf(mrow[mh][r] - mnew);
-        mrow[mh][r] = mnew;
-      }
-      float psum[4] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int nt = 0; nt < 4; ++nt)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float p = (sfrag[mh][nt][r] == -INFINITY)
-                        ? 0.0f
-                        : __expf(sfrag[mh][nt][r] - mrow[mh][r]);
-          sfrag[mh][nt][r] = p;
-          psum[r] += p;
+        for (int mt = 0; mt < 4; ++mt) {
+          f32x4 s = sfrag[mt][nq];
+          tmax = fmaxf(tmax, fmaxf(fmaxf(s[0], s[1]), fmaxf(s[2], s[3])));
         }
+        // row spread over lanes l15, l15+16, l15+32, l15+48
+        tmax = fmaxf(tmax, __shfl_xor(tmax, 16, WAVE));
+        tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
+        float mnew = fmaxf(mrow[nq], tmax);
+        alpha[nq] =
+            (mrow[nq] == -INFINITY) ? 0.0f : exp2f((mrow[nq] - mnew) * c);
+        mrow[nq] = mnew;
+        float psum = 0.f;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        psum[r] = group16_sum(psum[r]);
-        lrow[mh][r] = lrow[mh][r] * alpha[mh][r] + psum[r];
-      }
-      // stage P (swizzled [32 q][64 k] per-wave tile)
-      short* pl = p_lds[wid];
+        for (int mt = 0; mt < 4; ++mt) {
+          f32x4 s = sfrag[mt][nq];
+          f32x4 p;
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
-        int col = nt * 16 + l15;
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int row = mh * 16 + lg * 4 + r;
-          pl[swz(row, col, KVBLK)] = f2bf(sfrag[mh][nt][r]);
+          for (int r = 0; r < 4; ++r) {
+            p[r] = (s[r] == -INFINITY) ? 0.0f : exp2f((s[r] - mnew) * c);
+            psum += p[r];
+          }
+          // keys of this C tile are already in B-operand order: mt -> (kc =
+          // mt>>1, reg pair = (mt&1)*2 + {0,1})
+          pk[nq][mt >> 1][(mt & 1) * 2 + 0] = cvt_pk_bf16(p[0], p[1]);
+          pk[nq][mt >> 1][(mt & 1) * 2 + 1] = cvt_pk_bf16(p[2], p[3]);
         }
+        psum += __shfl_xor(psum, 16, WAVE);
+        psum += __shfl_xor(psum, 32, WAVE);
+        lrow[nq] = lrow[nq] * alpha[nq] + psum;
       }
-    }
-    }  // active
-    __builtin_amdgcn_s_barrier();
 
-    // ---- P A-frags + O accumulate ----
-    if (active) {
-    bf16x8 pa[2][2];
-#pragma unroll
-    for (int mh = 0; mh < 2; ++mh)
-#pragma unroll
-      for (int kc = 0; kc < 2; ++kc) {
-        int row = mh * 16 + l15;
-        int kk = kc * 32 + lg * 8;
-        pa[mh][kc] = *reinterpret_cast<const bf16x8*>(
-            p_lds[wid] + swz(row, kk, KVBLK));
-      }
-    bool rescale_done[2] = {false, false};
-#pragma unroll
-    for (int dt = 0; dt < 8; ++dt) {
-      if (dt >= nd) break;
-      bf16x8 vf[2];
-#pragma unroll
-      for (int kc = 0; kc < 2; ++kc) {
-        int d = dt * 16 + l15;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          int key = kc * 32 + lg * 8 + j;
-          vf[kc][j] = v_lds[swz(key, d, D)];
-        }
-      }
-#pragma unroll
-      for (int mh = 0; mh < 2; ++mh) {
-        f32x4 acc = oacc[mh][dt];
-#pragma unroll
-        for (int r = 0; r < 4; ++r) acc[r] *= alpha[mh][r];
-        acc = mfma16(pa[mh][0], vf[0], acc);
-        acc = mfma16(pa[mh][1], vf[1], acc);
-        oacc[mh][dt] = acc;
-      }
-    }
-    (void)rescale_done;
-    }  // active
-    __builtin_amdgcn_s_barrier();
-  }
-
-  // ---- epilogue ----
-#pragma unroll
-  for (int mh = 0; mh < 2; ++mh)
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      int qrow = q0 + mh * 16 + lg * 4 + r;
-      if (qrow >= S) continue;
-      float invl = (lrow[mh][r] > 0.f) ? 1.0f / lrow[mh][r] : 0.0f;
-      short* op = o + ((int64_t)b * S * Hq + (int64_t)qrow * Hq + h) * D;
+      // ---- O^T += mfma(V^T, P^T) ----
+      const short* vt = vt_lds[buf];
 #pragma unroll
       for (int dt = 0; dt < 8; ++dt) {
-        if (dt >= nd) break;
-        op[dt * 16 + l15] = f2bf(oacc[mh][dt][r] * invl);
+        if (dt >= nd16) break;
+        const int d = dt * 16 + l15;
+        bf16x8 va[2];
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc)
+          va[kc] = *reinterpret_cast<const bf16x8*>(
+              vt + vt_idx(d, kc * 32 + lg * 8));
+#pragma unroll
+        for (int nq = 0; nq < 2; ++nq) {
+          f32x4 acc = oacc[dt][nq];
+#pragma unroll
+          for (int r = 0; r < 4; ++r) acc[r] *= alpha[nq];
+          acc = mfma16(va[0], *reinterpret_cast<const bf16x8*>(&pk[nq][0][0]),
+                       acc);
+          acc = mfma16(va[1], *reinterpret_cast<const bf16x8*>(&pk[nq][1][0]),
+                       acc);
+          oacc[dt][nq] = acc;
+        }
       }
-      if (l15 == 0)
-        lse_out[((int64_t)b * Hq + h) * S + qrow] =
-            mrow[mh][r] + __logf(fmaxf(lrow[mh][r], 1e-30f));
+    }  // active
+    __syncthreads();
+  }
+
+  // ---- epilogue: lane holds qrow = l15, d = dt*16 + lg*4 + r ----
+#pragma unroll
+  for (int nq = 0; nq < 2; ++nq) {
+    const int qrow = q0 + nq * 16 + l15;
+    if (qrow >= S) continue;
+    const float invl = (lrow[nq] > 0.f) ? 1.0f / lrow[nq] : 0.0f;
+    short* op = o + ((int64_t)b * S * Hq + (int64_t)qrow * Hq + h) * D;
+#pragma unroll
+    for (int dt = 0; dt < 8; ++dt) {
+      if (dt >= nd16) break;
+      s16x4v outv;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) outv[r] = f2bf(oacc[dt][nq][r] * invl);
+      *reinterpret_cast<s16x4v*>(op + dt * 16 + lg * 4) = outv;
     }
+    if (lg == 0)
+      lse_out[((int64_t)b * Hq + h) * S + qrow] =
+          mrow[nq] * scale + __logf(fmaxf(lrow[nq], 1e-30f));
+  }
 }
 
 extern "C" {

@@ -89,3 +89,30 @@ __device__ __forceinline__ float block_reduce_max(float v, float* scratch) {
     hipError_t _e = (expr);                                                  \
     if (_e != hipSuccess) return _e;                                         \
   } while (0)
+
+// ---- attention kernel shared idioms (fwd v3 / bwd v2) ----
+// pack two f32 into one reg of 2 bf16 (no builtin on gfx950)
+__device__ __forceinline__ uint32_t cvt_pk_bf16(float lo, float hi) {
+  uint32_t r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+// transposed [D<=128][64] bf16 LDS image: index swizzled so the scalar
+// transposed staging writes spread banks while the per-row 8-col block
+// read stays 16B-contiguous.
+__device__ __forceinline__ int tr_idx(int d, int col) {
+  return d * 64 + (col ^ ((d & 7) << 3) ^ (((d >> 3) & 7) << 3));
+}
+
+// MFMA C-fragment "already-in-B/A-operand-order" row permutation: feeding
+// A-operand rows of tile mt in order perm16(mt, l15) makes C position
+// (mt, lg, r) hold source row (mt>>1)*32 + lg*8 + (mt&1)*4 + r, i.e. the
+// packed C rows are directly the k-dim layout (kc*32 + lg*8 + j) that the
+// next MFMA's A/B operand wants.
+__device__ __forceinline__ int perm16(int mt, int l15) {
+  return (mt >> 1) * 32 + (l15 >> 2) * 8 + (mt & 1) * 4 + (l15 & 3);
+}
+__device__ __forceinline__ int cpos16(int mt, int lg) {
+  return (mt >> 1) * 32 + lg * 8 + (mt & 1) * 4;
+}
