@@ -43,13 +43,6 @@ __device__ __forceinline__ f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
-// pack two f32 into one reg of 2 bf16 (no builtin on gfx950 — guide T12)
-__device__ __forceinline__ uint32_t cvt_pk_bf16(float lo, float hi) {
-  uint32_t r;
-  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
-  return r;
-}
-
 #define QBLK 128   // q rows per block (32 per wave)
 #define KVBLK 64   // keys per LDS tile
 #define LOG2E 1.4426950408889634f
