@@ -31,48 +31,59 @@ __global__ void __launch_bounds__(256) adamw_kernel(
     const bool pbf = d.p_is_bf16 != 0;
     const bool gbf = d.g_is_bf16 != 0;
     const float decay = 1.0f - lr * wd;
-    // vectorized main body: 4 elements per thread per iteration (16 B f32
-    // moments, 8/16 B params+grads — memory-bound, G13 vectorization)
-    const int64_t n4 = d.n / 4;
-    for (int64_t q = threadIdx.x; q < n4; q += blockDim.x) {
-      int64_t i = q * 4;
-      f32x4 g4, p4;
+    // vectorized main body: 8 elements per thread per iteration (2x 16 B
+    // f32 moment streams in flight — memory-bound, G13 vectorization +
+    // deeper MLP; update via rsqrt: p -= lr*m̂ * rsqrt-style reciprocal)
+    const int64_t n8 = d.n / 8;
+    for (int64_t qq = threadIdx.x; qq < n8; qq += blockDim.x) {
+      int64_t i = qq * 8;
+      f32x4 g4[2], p4[2], m4[2], v4[2];
       if (gbf) {
-        s16x4v gv = *reinterpret_cast<const s16x4v*>((const short*)d.g + i);
+        s16x8 gv = *reinterpret_cast<const s16x8*>((const short*)d.g + i);
 #pragma unroll
-        for (int j = 0; j < 4; ++j) g4[j] = bf2f(gv[j]);
+        for (int j = 0; j < 8; ++j) g4[j >> 2][j & 3] = bf2f(gv[j]);
       } else {
-        g4 = *reinterpret_cast<const f32x4*>((const float*)d.g + i);
+        g4[0] = *reinterpret_cast<const f32x4*>((const float*)d.g + i);
+        g4[1] = *reinterpret_cast<const f32x4*>((const float*)d.g + i + 4);
       }
       if (pbf) {
-        s16x4v pv = *reinterpret_cast<const s16x4v*>((const short*)d.p + i);
+        s16x8 pv = *reinterpret_cast<const s16x8*>((const short*)d.p + i);
 #pragma unroll
-        for (int j = 0; j < 4; ++j) p4[j] = bf2f(pv[j]);
+        for (int j = 0; j < 8; ++j) p4[j >> 2][j & 3] = bf2f(pv[j]);
       } else {
-        p4 = *reinterpret_cast<const f32x4*>((const float*)d.p + i);
+        p4[0] = *reinterpret_cast<const f32x4*>((const float*)d.p + i);
+        p4[1] = *reinterpret_cast<const f32x4*>((const float*)d.p + i + 4);
       }
-      f32x4 m4 = *reinterpret_cast<const f32x4*>(m + i);
-      f32x4 v4 = *reinterpret_cast<const f32x4*>(v + i);
+      m4[0] = *reinterpret_cast<const f32x4*>(m + i);
+      m4[1] = *reinterpret_cast<const f32x4*>(m + i + 4);
+      v4[0] = *reinterpret_cast<const f32x4*>(v + i);
+      v4[1] = *reinterpret_cast<const f32x4*>(v + i + 4);
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        m4[j] = beta1 * m4[j] + (1.0f - beta1) * g4[j];
-        v4[j] = beta2 * v4[j] + (1.0f - beta2) * g4[j] * g4[j];
-        p4[j] = p4[j] * decay -
-                lr * (m4[j] * inv_bc1) / (sqrtf(v4[j] * inv_bc2) + eps);
-      }
-      *reinterpret_cast<f32x4*>(m + i) = m4;
-      *reinterpret_cast<f32x4*>(v + i) = v4;
+      for (int h = 0; h < 2; ++h)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          m4[h][j] = beta1 * m4[h][j] + (1.0f - beta1) * g4[h][j];
+          v4[h][j] = beta2 * v4[h][j] + (1.0f - beta2) * g4[h][j] * g4[h][j];
+          p4[h][j] = p4[h][j] * decay -
+                     lr * (m4[h][j] * inv_bc1) /
+                         (sqrtf(v4[h][j] * inv_bc2) + eps);
+        }
+      *reinterpret_cast<f32x4*>(m + i) = m4[0];
+      *reinterpret_cast<f32x4*>(m + i + 4) = m4[1];
+      *reinterpret_cast<f32x4*>(v + i) = v4[0];
+      *reinterpret_cast<f32x4*>(v + i + 4) = v4[1];
       if (pbf) {
-        s16x4v pv;
+        s16x8 pv;
 #pragma unroll
-        for (int j = 0; j < 4; ++j) pv[j] = f2bf(p4[j]);
-        *reinterpret_cast<s16x4v*>((short*)d.p + i) = pv;
+        for (int j = 0; j < 8; ++j) pv[j] = f2bf(p4[j >> 2][j & 3]);
+        *reinterpret_cast<s16x8*>((short*)d.p + i) = pv;
       } else {
-        *reinterpret_cast<f32x4*>((float*)d.p + i) = p4;
+        *reinterpret_cast<f32x4*>((float*)d.p + i) = p4[0];
+        *reinterpret_cast<f32x4*>((float*)d.p + i + 4) = p4[1];
       }
     }
     // scalar tail
-    for (int64_t i = n4 * 4 + threadIdx.x; i < d.n; i += blockDim.x) {
+    for (int64_t i = n8 * 8 + threadIdx.x; i < d.n; i += blockDim.x) {
       float g = gbf ? bf2f(((const short*)d.g)[i]) : ((const float*)d.g)[i];
       float p = pbf ? bf2f(((const short*)d.p)[i]) : ((const float*)d.p)[i];
       float mi = m[i] = beta1 * m[i] + (1.0f - beta1) * g;

@@ -33,9 +33,10 @@ def main():
     print(f"[rank={rank}/{world} restart={os.environ.get('TORCHELASTIC_RESTART_COUNT', 0)}] "
           f"resuming at iteration {state['iteration']}")
 
+    fail_prob = float(os.environ.get("TOY_FAIL_PROB", "0.05"))
     for it in range(state["iteration"], 20):
         time.sleep(0.1)  # "training"
-        if random.random() < 0.05:
+        if random.random() < fail_prob:
             raise RuntimeError(f"rank {rank} simulated failure at iter {it}")
         state["iteration"] = it + 1
         dist.barrier()
