@@ -320,7 +320,10 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
   const int l15 = lane & 15;
   const int lg = lane >> 4;
 
-  const int q0 = qtile * 128 + wid * 32;  // this wave's 32 q rows
+  // this wave's 32 q rows as two 16-row groups interleaved across the
+  // block (rows w*16 and w*16+64) — causal activity is then uniform
+  // across waves (no per-tile barrier idling); see attention_fwd.hip
+  const int rowb[2] = {qtile * 128 + wid * 16, qtile * 128 + 64 + wid * 16};
   const int nd16 = D >> 4, nkc = D >> 5;
   const int64_t strideS_q = (int64_t)Hq * D;
   const int64_t strideS_kv = (int64_t)Hkv * D;
@@ -336,7 +339,7 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
   float lse_r[2], del_r[2];
 #pragma unroll
   for (int nq = 0; nq < 2; ++nq) {
-    int qrow = q0 + nq * 16 + l15;
+    int qrow = rowb[nq] + l15;
     int qrc = qrow < S ? qrow : S - 1;
     const short* qp = qb + (int64_t)qrc * strideS_q + lg * 8;
     const short* dop = dob + (int64_t)qrc * strideS_q + lg * 8;
@@ -389,12 +392,14 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
     const int buf = t & 1;
     if (t + 1 < ntiles) stage_k((t + 1) * 64, buf ^ 1);
 
-    const bool active = (kv0 <= q0 + 31);
-    if (active) {
+    // group 0 (lower rows) drops out at the final diagonal tiles; group 1
+    // is live for every tile — uniform across waves (no barrier idling)
+    const bool act0 = kv0 <= rowb[0] + 15;
+    {
       // ---- per mt: S^T = mfma(K_perm, Q), dP^T = mfma(V_perm, dO), then
       // immediately exp/pack dS into the dQ MFMA's A operand (keys in
       // kc*32+lg*8+j order via perm16) — keeps only one mt of S/dP live ----
-      const bool diag = (kv0 + 63 > q0) || (kv_end < kv0 + 64);
+      const bool smask = (kv_end < kv0 + 64);
       uint32_t pk_ds[2][2][4];  // [nq][kc][4]
 #pragma unroll
       for (int mt = 0; mt < 4; ++mt) {
@@ -409,16 +414,20 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
           if (c < nkc) {
             bf16x8 ka = *reinterpret_cast<const bf16x8*>(kp + c * 32);
             bf16x8 va = *reinterpret_cast<const bf16x8*>(vp + c * 32);
-            s0 = mfma16b(ka, qf[0][c], s0);
+            if (act0) {
+              s0 = mfma16b(ka, qf[0][c], s0);
+              d0 = mfma16b(va, dof[0][c], d0);
+            }
             s1 = mfma16b(ka, qf[1][c], s1);
-            d0 = mfma16b(va, dof[0][c], d0);
             d1 = mfma16b(va, dof[1][c], d1);
           }
         const int koff = cpos16(mt, lg);
         const int kc = mt >> 1, rp = (mt & 1) * 2;
 #pragma unroll
         for (int nq = 0; nq < 2; ++nq) {
-          const int qrow = q0 + nq * 16 + l15;
+          if (nq == 0 && !act0) continue;
+          const int qrow = rowb[nq] + l15;
+          const bool diag = (kv0 + 63 > rowb[nq]) || smask;
           const f32x4 sv = nq ? s1 : s0;
           const f32x4 dv = nq ? d1 : d0;
           float ds[4];
@@ -447,14 +456,16 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
         for (int kc = 0; kc < 2; ++kc) {
           bf16x8 kfr = *reinterpret_cast<const bf16x8*>(
               ktl + tr_idx(d, kc * 32 + lg * 8));
-#pragma unroll
-          for (int nq = 0; nq < 2; ++nq)
-            dqacc[dt][nq] = mfma16b(
-                *reinterpret_cast<const bf16x8*>(&pk_ds[nq][kc][0]), kfr,
-                dqacc[dt][nq]);
+          if (act0)
+            dqacc[dt][0] = mfma16b(
+                *reinterpret_cast<const bf16x8*>(&pk_ds[0][kc][0]), kfr,
+                dqacc[dt][0]);
+          dqacc[dt][1] = mfma16b(
+              *reinterpret_cast<const bf16x8*>(&pk_ds[1][kc][0]), kfr,
+              dqacc[dt][1]);
         }
       }
-    }  // active
+    }
     __syncthreads();
   }
 
@@ -463,7 +474,7 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
   for (int nq = 0; nq < 2; ++nq) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      int qrow = q0 + nq * 16 + lg * 4 + r;
+      int qrow = rowb[nq] + lg * 4 + r;
       if (qrow >= S) continue;
       short* dqp = dq + ((int64_t)b * S * Hq + (int64_t)qrow * Hq + h) * D;
 #pragma unroll

@@ -93,7 +93,11 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
   const int l15 = lane & 15;
   const int lg = lane >> 4;
 
-  const int q0 = qtile * QBLK + wid * 32;  // this wave's 32 q rows
+  // this wave's 32 q rows, as two 16-row groups INTERLEAVED across the
+  // block (rows w*16 and w*16+64): every wave then touches both halves of
+  // the q range, so causal tile-skipping is uniform across waves and no
+  // wave idles at the per-tile barrier (guide: wave load balance)
+  const int rowb[2] = {qtile * QBLK + wid * 16, qtile * QBLK + 64 + wid * 16};
   const int nd16 = D >> 4;                 // 16-d tiles (8 for D=128)
   const int nkc = D >> 5;                  // 32-d MFMA K chunks
   const int64_t strideS_q = (int64_t)Hq * D;
@@ -106,7 +110,7 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
   bf16x8 qf[2][4];
 #pragma unroll
   for (int nq = 0; nq < 2; ++nq) {
-    int qrow = q0 + nq * 16 + l15;
+    int qrow = rowb[nq] + l15;
     if (qrow >= S) qrow = S - 1;
     const short* qp = qb + (int64_t)qrow * strideS_q + lg * 8;
 #pragma unroll
@@ -142,6 +146,7 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
   const int kv_end = min(S, qtile * QBLK + QBLK);
   const int ntiles = (kv_end + KVBLK - 1) / KVBLK;
   const float c = scale * LOG2E;
+  (void)hkv;
 
   // ---- staging: thread stages V rows (key = tid&63), 16B of d each ----
   const int skey = tid & 63;
@@ -175,107 +180,112 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
     // MFMA stream — T14 spirit; ds_writes don't touch the compute buffer)
     if (t + 1 < ntiles) stage_v((t + 1) * KVBLK, buf ^ 1);
 
-    const bool active = (kv0 <= q0 + 31);
-    if (active) {
-      // ---- S^T = mfma(K, Q): [mt 4][nq 2] C frags ----
-      f32x4 sfrag[4][2];
-#pragma unroll
-      for (int mt = 0; mt < 4; ++mt) {
-        int keyg = kv0 + kperm[mt];
-        if (keyg >= S) keyg = S - 1;
-        const short* kp = kb + (int64_t)keyg * strideS_kv + lg * 8;
-        f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-        for (int kc = 0; kc < 4; ++kc)
-          if (kc < nkc) {
-            bf16x8 kf = *reinterpret_cast<const bf16x8*>(kp + kc * 32);
-            acc0 = mfma16(kf, qf[0][kc], acc0);
-            acc1 = mfma16(kf, qf[1][kc], acc1);
-          }
-        sfrag[mt][0] = acc0;
-        sfrag[mt][1] = acc1;
-      }
+    // per-16-row-group activity (wave-uniform): group 1 (upper rows) is
+    // live for every tile of this block; group 0 drops out only at the
+    // final diagonal tiles — uniform across waves, so no barrier idling
+    const bool act0 = kv0 <= rowb[0] + 15;
 
-      // ---- causal mask + online softmax (rows lane-local) ----
-      const bool need_mask = (kv0 + KVBLK - 1) > q0 || kv_end < kv0 + KVBLK;
-      uint32_t pk[2][2][4];  // [nq][kc][4 regs of 2 bf16]
-      float alpha[2];
+    // ---- S^T = mfma(K, Q): [mt 4][nq 2] C frags ----
+    f32x4 sfrag[4][2];
 #pragma unroll
-      for (int nq = 0; nq < 2; ++nq) {
-        const int qrow = q0 + nq * 16 + l15;
-        if (need_mask) {
+    for (int mt = 0; mt < 4; ++mt) {
+      int keyg = kv0 + kperm[mt];
+      if (keyg >= S) keyg = S - 1;
+      const short* kp = kb + (int64_t)keyg * strideS_kv + lg * 8;
+      f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-          for (int mt = 0; mt < 4; ++mt)
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              int key = kv0 + ckey[mt] + r;
-              if (key > qrow || key >= S) sfrag[mt][nq][r] = -INFINITY;
-            }
+      for (int kc = 0; kc < 4; ++kc)
+        if (kc < nkc) {
+          bf16x8 kf = *reinterpret_cast<const bf16x8*>(kp + kc * 32);
+          if (act0) acc0 = mfma16(kf, qf[0][kc], acc0);
+          acc1 = mfma16(kf, qf[1][kc], acc1);
         }
-        float tmax = -INFINITY;
+      sfrag[mt][0] = acc0;
+      sfrag[mt][1] = acc1;
+    }
+
+    // ---- causal mask + online softmax (rows lane-local) ----
+    uint32_t pk[2][2][4];  // [nq][kc][4 regs of 2 bf16]
+    float alpha[2];
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt) {
-          f32x4 s = sfrag[mt][nq];
-          tmax = fmaxf(tmax, fmaxf(fmaxf(s[0], s[1]), fmaxf(s[2], s[3])));
-        }
-        // row spread over lanes l15, l15+16, l15+32, l15+48
-        tmax = fmaxf(tmax, __shfl_xor(tmax, 16, WAVE));
-        tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
-        float mnew = fmaxf(mrow[nq], tmax);
-        alpha[nq] =
-            (mrow[nq] == -INFINITY) ? 0.0f : exp2f((mrow[nq] - mnew) * c);
-        mrow[nq] = mnew;
-        float psum = 0.f;
+    for (int nq = 0; nq < 2; ++nq) {
+      if (nq == 0 && !act0) continue;
+      const int qrow = rowb[nq] + l15;
+      const bool need_mask =
+          (kv0 + KVBLK - 1) > rowb[nq] || kv_end < kv0 + KVBLK;
+      if (need_mask) {
 #pragma unroll
-        for (int mt = 0; mt < 4; ++mt) {
-          f32x4 s = sfrag[mt][nq];
-          f32x4 p;
+        for (int mt = 0; mt < 4; ++mt)
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
-            p[r] = (s[r] == -INFINITY) ? 0.0f : exp2f((s[r] - mnew) * c);
-            psum += p[r];
+            int key = kv0 + ckey[mt] + r;
+            if (key > qrow || key >= S) sfrag[mt][nq][r] = -INFINITY;
           }
-          // keys of this C tile are already in B-operand order: mt -> (kc =
-          // mt>>1, reg pair = (mt&1)*2 + {0,1})
-          pk[nq][mt >> 1][(mt & 1) * 2 + 0] = cvt_pk_bf16(p[0], p[1]);
-          pk[nq][mt >> 1][(mt & 1) * 2 + 1] = cvt_pk_bf16(p[2], p[3]);
-        }
-        psum += __shfl_xor(psum, 16, WAVE);
-        psum += __shfl_xor(psum, 32, WAVE);
-        lrow[nq] = lrow[nq] * alpha[nq] + psum;
       }
+      float tmax = -INFINITY;
+#pragma unroll
+      for (int mt = 0; mt < 4; ++mt) {
+        f32x4 s = sfrag[mt][nq];
+        tmax = fmaxf(tmax, fmaxf(fmaxf(s[0], s[1]), fmaxf(s[2], s[3])));
+      }
+      // row spread over lanes l15, l15+16, l15+32, l15+48
+      tmax = fmaxf(tmax, __shfl_xor(tmax, 16, WAVE));
+      tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
+      float mnew = fmaxf(mrow[nq], tmax);
+      alpha[nq] =
+          (mrow[nq] == -INFINITY) ? 0.0f : exp2f((mrow[nq] - mnew) * c);
+      mrow[nq] = mnew;
+      float psum = 0.f;
+#pragma unroll
+      for (int mt = 0; mt < 4; ++mt) {
+        f32x4 s = sfrag[mt][nq];
+        f32x4 p;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          p[r] = (s[r] == -INFINITY) ? 0.0f : exp2f((s[r] - mnew) * c);
+          psum += p[r];
+        }
+        // keys of this C tile are already in B-operand order: mt -> (kc =
+        // mt>>1, reg pair = (mt&1)*2 + {0,1})
+        pk[nq][mt >> 1][(mt & 1) * 2 + 0] = cvt_pk_bf16(p[0], p[1]);
+        pk[nq][mt >> 1][(mt & 1) * 2 + 1] = cvt_pk_bf16(p[2], p[3]);
+      }
+      psum += __shfl_xor(psum, 16, WAVE);
+      psum += __shfl_xor(psum, 32, WAVE);
+      lrow[nq] = lrow[nq] * alpha[nq] + psum;
+    }
 
-      // ---- O^T += mfma(V^T, P^T) ----
-      const short* vt = vt_lds[buf];
+    // ---- O^T += mfma(V^T, P^T) ----
+    const short* vt = vt_lds[buf];
 #pragma unroll
-      for (int dt = 0; dt < 8; ++dt) {
-        if (dt >= nd16) break;
-        const int d = dt * 16 + l15;
-        bf16x8 va[2];
+    for (int dt = 0; dt < 8; ++dt) {
+      if (dt >= nd16) break;
+      const int d = dt * 16 + l15;
+      bf16x8 va[2];
 #pragma unroll
-        for (int kc = 0; kc < 2; ++kc)
-          va[kc] = *reinterpret_cast<const bf16x8*>(
-              vt + vt_idx(d, kc * 32 + lg * 8));
+      for (int kc = 0; kc < 2; ++kc)
+        va[kc] = *reinterpret_cast<const bf16x8*>(
+            vt + vt_idx(d, kc * 32 + lg * 8));
 #pragma unroll
-        for (int nq = 0; nq < 2; ++nq) {
-          f32x4 acc = oacc[dt][nq];
+      for (int nq = 0; nq < 2; ++nq) {
+        if (nq == 0 && !act0) continue;
+        f32x4 acc = oacc[dt][nq];
 #pragma unroll
-          for (int r = 0; r < 4; ++r) acc[r] *= alpha[nq];
-          acc = mfma16(va[0], *reinterpret_cast<const bf16x8*>(&pk[nq][0][0]),
-                       acc);
-          acc = mfma16(va[1], *reinterpret_cast<const bf16x8*>(&pk[nq][1][0]),
-                       acc);
-          oacc[dt][nq] = acc;
-        }
+        for (int r = 0; r < 4; ++r) acc[r] *= alpha[nq];
+        acc = mfma16(va[0], *reinterpret_cast<const bf16x8*>(&pk[nq][0][0]),
+                     acc);
+        acc = mfma16(va[1], *reinterpret_cast<const bf16x8*>(&pk[nq][1][0]),
+                     acc);
+        oacc[dt][nq] = acc;
       }
-    }  // active
+    }
     __syncthreads();
   }
 
   // ---- epilogue: lane holds qrow = l15, d = dt*16 + lg*4 + r ----
 #pragma unroll
   for (int nq = 0; nq < 2; ++nq) {
-    const int qrow = q0 + nq * 16 + l15;
+    const int qrow = rowb[nq] + l15;
     if (qrow >= S) continue;
     const float invl = (lrow[nq] > 0.f) ? 1.0f / lrow[nq] : 0.0f;
     short* op = o + ((int64_t)b * S * Hq + (int64_t)qrow * Hq + h) * D;
