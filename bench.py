@@ -43,6 +43,26 @@ def parse_args():
     return p.parse_args()
 
 
+def _maybe_load_tunableop(model_name: str, device):
+    """Load a committed hipBLASLt/rocBLAS TunableOp result file (offline
+    GEMM autotuning for this model's shapes) if one exists in profiles/.
+    Tuning itself is done offline (see profiles/README); here we only READ
+    the chosen solutions — no runtime tuning overhead."""
+    if device.type != "cuda" or os.environ.get("PYTORCH_TUNABLEOP_TUNING"):
+        return
+    f = Path(__file__).parent / "profiles" / f"tunableop_{model_name}.csv"
+    if not f.exists():
+        return
+    try:
+        import torch.cuda.tunable as tunable
+
+        tunable.enable(True)
+        tunable.tuning_enable(False)
+        tunable.read_file(str(f))
+    except Exception as e:  # noqa: BLE001 - perf feature, never fatal
+        print(f"tunableop load failed: {e}", file=sys.stderr)
+
+
 def main():
     args = parse_args()
     import torch.distributed as dist
@@ -65,6 +85,8 @@ def main():
         from distributed_training_guide_amd.parallel.pg import init_distributed
 
         init_distributed(device)
+
+    _maybe_load_tunableop(args.model, device)
 
     torch.manual_seed(1234 + rank)
     config = get_config(args.model)
