@@ -1,5 +1,6 @@
-// Causal flash-attention forward for gfx950 — v3: swapped-operand S^T MFMA
-// with in-register P, direct-from-L2 K, transposed double-buffered V LDS.
+// Causal flash-attention forward for gfx950 — v5: swapped-operand S^T MFMA
+// with in-register P, double-buffered LDS K/V tiles prefetched a full tile
+// ahead (T14), one barrier per tile.
 //
 // Replaces the reference's flash-attn-2 dependency
 // (attn_implementation="flash_attention_2", 05:93 / 06:73 / 07:71 —
@@ -17,11 +18,14 @@
 //     B-operand layout that P^T needs for the P·V MFMA. P therefore never
 //     leaves registers: exp -> v_cvt_pk_bf16_f32 -> B fragment. No P LDS
 //     round-trip, no block barrier between QK^T and P·V.
-//   * K is NOT staged in LDS: at training shapes the K tile is L1/L2
-//     resident (guide §5 mistake 7) and the A-fragment read of K is a
-//     clean bf16x8 global load (8 consecutive d per lane). The XCD-aware
-//     block remap below keeps all blocks of one (b, kv-head) on one XCD's
-//     L2.
+//   * K and V are staged into double-buffered LDS, with the global loads
+//     for tile t+1 issued at the TOP of tile t's compute: the ~200-900cy
+//     global latency lands behind a full tile of MFMA (T14), and the
+//     A-fragment reads become ~50cy ds_read_b128 the compiler interleaves
+//     with the MFMA stream (PMC before this change: 67% SQ_WAIT_ANY on the
+//     direct-global K path). K image row-major 16B-slot-swizzled
+//     (conflict-free writes AND reads); the XCD-aware remap keeps a
+//     (b,kv-head) group's K/V on one XCD's L2.
 //   * V is staged TRANSPOSED into double-buffered LDS (v^T[d][key], key
 //     index XOR-swizzled): the P·V MFMA's A operand needs 8 consecutive
 //     keys per lane, which only a transposed image can serve as
@@ -60,6 +64,7 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
     float* __restrict__ lse_out, int B, int S, int Hq, int Hkv, int D,
     float scale) {
   __shared__ short vt_lds[2][128 * KVBLK];
+  __shared__ short k_lds[2][KVBLK * 128];  // row-major [key][d], slot-swizzled
 
   const int ntq = (S + QBLK - 1) / QBLK;
   const int gqa = Hq / Hkv;
@@ -148,10 +153,16 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
   const float c = scale * LOG2E;
   (void)hkv;
 
-  // ---- staging: thread stages V rows (key = tid&63), 16B of d each ----
+  // ---- staging ----
+  // V: thread stages V rows transposed (key = tid&63, 16B of d per slot).
+  // K: staged ROW-major [key][d] with the 16B-slot XOR swizzle (16 lanes
+  //    cover one key row -> conflict-free b128 writes, coalesced reads);
+  //    both tiles are loaded ONE TILE AHEAD into the other buffer so the
+  //    global latency lands behind a full tile of MFMA (T14).
   const int skey = tid & 63;
   const int sslot0 = (tid >> 6) * 2;  // wave w covers d slots {2w,2w+1,8+2w,8+2w+1}
-  auto stage_v = [&](int kv0, int buf) {
+  const int nslot = D >> 3;
+  auto stage_kv = [&](int kv0, int buf) {
     int keyg = kv0 + skey;
     if (keyg >= S) keyg = S - 1;
     const short* vp = vb + (int64_t)keyg * strideS_kv;
@@ -168,9 +179,22 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
         for (int i = 0; i < 8; ++i) dst[vt_idx(d0 + i, skey)] = vec[i];
       }
     }
+    short* kdst = k_lds[buf];
+#pragma unroll
+    for (int vv = 0; vv < 4; ++vv) {
+      int vecid = vv * 256 + tid;
+      int key = vecid / 16, slot = vecid & 15;
+      if (slot >= nslot || key >= KVBLK) continue;
+      int kg = kv0 + key;
+      if (kg >= S) kg = S - 1;
+      bf16x8 vec = *reinterpret_cast<const bf16x8*>(
+          kb + (int64_t)kg * strideS_kv + slot * 8);
+      *reinterpret_cast<bf16x8*>(
+          kdst + key * D + ((slot ^ (key & 7)) & (nslot - 1)) * 8) = vec;
+    }
   };
 
-  stage_v(0, 0);
+  stage_kv(0, 0);
   __syncthreads();
 
   for (int t = 0; t < ntiles; ++t) {
@@ -178,25 +202,26 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
     const int buf = t & 1;
     // stage next tile into the other buffer (global loads issue before the
     // MFMA stream — T14 spirit; ds_writes don't touch the compute buffer)
-    if (t + 1 < ntiles) stage_v((t + 1) * KVBLK, buf ^ 1);
+    if (t + 1 < ntiles) stage_kv((t + 1) * KVBLK, buf ^ 1);
 
     // per-16-row-group activity (wave-uniform): group 1 (upper rows) is
     // live for every tile of this block; group 0 drops out only at the
     // final diagonal tiles — uniform across waves, so no barrier idling
     const bool act0 = kv0 <= rowb[0] + 15;
 
-    // ---- S^T = mfma(K, Q): [mt 4][nq 2] C frags ----
+    // ---- S^T = mfma(K, Q): [mt 4][nq 2] C frags; K A-frags from the
+    // row-major staged tile (ds_read_b128, ~50cyc, hidden by MFMA) ----
+    const short* kl = k_lds[buf];
     f32x4 sfrag[4][2];
 #pragma unroll
     for (int mt = 0; mt < 4; ++mt) {
-      int keyg = kv0 + kperm[mt];
-      if (keyg >= S) keyg = S - 1;
-      const short* kp = kb + (int64_t)keyg * strideS_kv + lg * 8;
+      const int kr = kperm[mt];  // local key row (permuted)
       f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int kc = 0; kc < 4; ++kc)
         if (kc < nkc) {
-          bf16x8 kf = *reinterpret_cast<const bf16x8*>(kp + kc * 32);
+          bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+              kl + kr * D + (((kc * 4 + lg) ^ (kr & 7)) & (nslot - 1)) * 8);
           if (act0) acc0 = mfma16(kf, qf[0][kc], acc0);
           acc1 = mfma16(kf, qf[1][kc], acc1);
         }
