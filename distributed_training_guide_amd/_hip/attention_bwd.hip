@@ -72,10 +72,12 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dk, short* __restrict__ dv, int B, int S, int Hq,
     int Hkv, int D, float scale) {
-  // transposed double-buffered staging of the 64-q-row tile:
-  // q^T[d][qrow], dO^T[d][qrow]; plus the tile's lse/delta rows
-  __shared__ short qt_lds[2][128 * 64];
-  __shared__ short dot_lds[2][128 * 64];
+  // double-buffered ROW-major staging of the 64-q-row tile (q[qrow][d],
+  // dO[qrow][d], rm_idx-swizzled): serves BOTH the Q/dO A-fragments
+  // (b128 row reads, replacing per-tile global gathers) and the
+  // dV/dK B-fragments (tr16 hardware-transpose reads); plus lse/delta rows
+  __shared__ short q_lds[2][64 * 128];
+  __shared__ short do_lds[2][64 * 128];
   __shared__ float lse_lds[2][64];
   __shared__ float del_lds[2][64];
 
@@ -142,35 +144,28 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
   const int nqt = (S - qstart + 63) / 64;
   const int niter = group * nqt;
 
-  // staging: thread stages one q row (qrow = qt + (tid&63)), 16B of d per
-  // slot; wave w covers d-slots {2w, 2w+1, 8+2w, 8+2w+1}
-  const int srow = tid & 63;
-  const int sslot0 = wid * 2;
+  // staging: 16 lanes per q row, vectorized b128 writes (rm_idx swizzle:
+  // conflict-free), one (h, q-tile) AHEAD so global latency hides behind
+  // a full tile of MFMA
+  const int nslot = D >> 3;
   auto stage = [&](int it, int buf) {
     const int hh = hkv * group + it / nqt;
     const int qt = qstart + (it % nqt) * 64;
-    int qrow = qt + srow;
-    if (qrow >= S) qrow = S - 1;
-    const short* qp = q + ((int64_t)b * S * Hq + hh) * D + qrow * strideS_q;
-    const short* dop =
-        dout + ((int64_t)b * S * Hq + hh) * D + qrow * strideS_q;
-    short* qdst = qt_lds[buf];
-    short* ddst = dot_lds[buf];
+    const short* qbase = q + ((int64_t)b * S * Hq + hh) * D;
+    const short* dobase = dout + ((int64_t)b * S * Hq + hh) * D;
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
-#pragma unroll
-      for (int ss = 0; ss < 2; ++ss) {
-        int slot = half * 8 + sslot0 + ss;
-        if (slot * 8 >= D) break;
-        bf16x8 qv = *reinterpret_cast<const bf16x8*>(qp + slot * 8);
-        bf16x8 dv = *reinterpret_cast<const bf16x8*>(dop + slot * 8);
-        int d0 = slot * 8;
-#pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          qdst[tr_idx(d0 + i, srow)] = qv[i];
-          ddst[tr_idx(d0 + i, srow)] = dv[i];
-        }
-      }
+    for (int vv = 0; vv < 4; ++vv) {
+      int vecid = vv * 256 + tid;
+      int row = vecid / 16, slot = vecid & 15;
+      if (slot >= nslot || row >= 64) continue;
+      int qrow = qt + row;
+      if (qrow >= S) qrow = S - 1;
+      bf16x8 qv = *reinterpret_cast<const bf16x8*>(
+          qbase + (int64_t)qrow * strideS_q + slot * 8);
+      bf16x8 dv = *reinterpret_cast<const bf16x8*>(
+          dobase + (int64_t)qrow * strideS_q + slot * 8);
+      *reinterpret_cast<bf16x8*>(q_lds[buf] + rm_idx(row, slot * 8, D)) = qv;
+      *reinterpret_cast<bf16x8*>(do_lds[buf] + rm_idx(row, slot * 8, D)) = dv;
     }
     if (tid < 64) {
       const float* lseb = lse + ((int64_t)b * Hq + hh) * S;
@@ -190,26 +185,24 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
     const int buf = it & 1;
     if (it + 1 < niter) stage(it + 1, buf ^ 1);
 
-    const int hh = hkv * group + it / nqt;
     const int qt = qstart + (it % nqt) * 64;
-    const short* qb = q + ((int64_t)b * S * Hq + hh) * D;
-    const short* dob = dout + ((int64_t)b * S * Hq + hh) * D;
+    const short* ql = q_lds[buf];
+    const short* dol = do_lds[buf];
 
     // ---- S[mt], dP[mt]: C[m=qrow(perm)][n=key]; A = Q/dO rows fed in
-    // perm16 order (direct global loads), B = kf/vf ----
+    // perm16 order (b128 LDS reads), B = kf/vf ----
     f32x4 sfrag[4], dpfrag[4];
 #pragma unroll
     for (int mt = 0; mt < 4; ++mt) {
-      int qrow = qt + perm16(mt, l15);
-      int qrc = qrow < S ? qrow : S - 1;
-      const short* qp = qb + (int64_t)qrc * strideS_q + lg * 8;
-      const short* dop = dob + (int64_t)qrc * strideS_q + lg * 8;
+      const int qr = perm16(mt, l15);  // local row in the staged tile
       f32x4 sa = {0.f, 0.f, 0.f, 0.f}, da = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int c = 0; c < 4; ++c)
         if (c < nkc) {
-          bf16x8 qa = *reinterpret_cast<const bf16x8*>(qp + c * 32);
-          bf16x8 doa = *reinterpret_cast<const bf16x8*>(dop + c * 32);
+          bf16x8 qa = *reinterpret_cast<const bf16x8*>(
+              ql + rm_idx(qr, c * 32 + lg * 8, D));
+          bf16x8 doa = *reinterpret_cast<const bf16x8*>(
+              dol + rm_idx(qr, c * 32 + lg * 8, D));
           sa = mfma16b(qa, kf[c], sa);
           da = mfma16b(doa, vf[c], da);
         }
@@ -221,7 +214,7 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
     // A-operand order (qrow k-dim = kc*32+lg*8+j via perm16) ----
     const int key = kv0 + l15;  // this lane's key (C n-position)
     const bool diag = (qt < kt * 64 + 64) || (qt + 63 >= S);
-    uint32_t pk_p[2][4], pk_ds[2][4];
+    u32x4 pk_p[2], pk_ds[2];
 #pragma unroll
     for (int mt = 0; mt < 4; ++mt) {
       const int qoff = cpos16(mt, lg);
@@ -242,23 +235,19 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
       pk_ds[kc][rp + 1] = cvt_pk_bf16(ds[2], ds[3]);
     }
 
-    // ---- dV += P^T dO ; dK += dS^T Q (B operands from transposed LDS) ----
-    const short* qtl = qt_lds[buf];
-    const short* dtl = dot_lds[buf];
+    // ---- dV += P^T dO ; dK += dS^T Q (B operands via tr16 transpose
+    // reads from the same row-major tiles) ----
 #pragma unroll
     for (int dt = 0; dt < 8; ++dt) {
       if (dt >= nd16) break;
-      const int d = dt * 16 + l15;
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
-        bf16x8 dofr = *reinterpret_cast<const bf16x8*>(
-            dtl + tr_idx(d, kc * 32 + lg * 8));
-        bf16x8 qfr = *reinterpret_cast<const bf16x8*>(
-            qtl + tr_idx(d, kc * 32 + lg * 8));
-        dvacc[dt] = mfma16b(*reinterpret_cast<const bf16x8*>(&pk_p[kc][0]),
-                            dofr, dvacc[dt]);
-        dkacc[dt] = mfma16b(*reinterpret_cast<const bf16x8*>(&pk_ds[kc][0]),
-                            qfr, dkacc[dt]);
+        bf16x8 dofr = tr16_frag(dol, kc * 32 + lg * 8, dt * 16, D, l15);
+        bf16x8 qfr = tr16_frag(ql, kc * 32 + lg * 8, dt * 16, D, l15);
+        dvacc[dt] = mfma16b(__builtin_bit_cast(bf16x8, pk_p[kc]), dofr,
+                            dvacc[dt]);
+        dkacc[dt] = mfma16b(__builtin_bit_cast(bf16x8, pk_ds[kc]), qfr,
+                            dkacc[dt]);
       }
     }
     __syncthreads();
@@ -288,8 +277,11 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dq, int B, int S, int Hq, int Hkv, int D,
     float scale) {
-  // transposed double-buffered K^T[d][key] staging
-  __shared__ short kt_lds[2][128 * 64];
+  // double-buffered ROW-major K and V tiles (rm_idx-swizzled): K serves
+  // the S^T A-fragments (b128 row reads) AND the dQ B-fragments (tr16
+  // transpose reads); V serves the dP^T A-fragments
+  __shared__ short k_lds[2][64 * 128];
+  __shared__ short v_lds[2][64 * 128];
 
   const int ntq = (S + 127) / 128;
   const int gqa = Hq / Hkv;
@@ -363,24 +355,21 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
   const int kv_end = min(S, qtile * 128 + 128);
   const int ntiles = (kv_end + 63) / 64;
 
-  const int srow = tid & 63;
-  const int sslot0 = wid * 2;
+  const int nslot = D >> 3;
   auto stage_k = [&](int kv0s, int buf) {
-    int keyg = kv0s + srow;
-    if (keyg >= S) keyg = S - 1;
-    const short* kp = kb + (int64_t)keyg * strideS_kv;
-    short* dst = kt_lds[buf];
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
-#pragma unroll
-      for (int ss = 0; ss < 2; ++ss) {
-        int slot = half * 8 + sslot0 + ss;
-        if (slot * 8 >= D) break;
-        bf16x8 vec = *reinterpret_cast<const bf16x8*>(kp + slot * 8);
-        int d0 = slot * 8;
-#pragma unroll
-        for (int i = 0; i < 8; ++i) dst[tr_idx(d0 + i, srow)] = vec[i];
-      }
+    for (int vv = 0; vv < 4; ++vv) {
+      int vecid = vv * 256 + tid;
+      int key = vecid / 16, slot = vecid & 15;
+      if (slot >= nslot || key >= 64) continue;
+      int kg = kv0s + key;
+      if (kg >= S) kg = S - 1;
+      bf16x8 kv_ = *reinterpret_cast<const bf16x8*>(
+          kb + (int64_t)kg * strideS_kv + slot * 8);
+      bf16x8 vv_ = *reinterpret_cast<const bf16x8*>(
+          vb + (int64_t)kg * strideS_kv + slot * 8);
+      *reinterpret_cast<bf16x8*>(k_lds[buf] + rm_idx(key, slot * 8, D)) = kv_;
+      *reinterpret_cast<bf16x8*>(v_lds[buf] + rm_idx(key, slot * 8, D)) = vv_;
     }
   };
 
@@ -394,30 +383,33 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
 
     // group 0 (lower rows) drops out at the final diagonal tiles; group 1
     // is live for every tile — uniform across waves (no barrier idling)
-    const bool act0 = kv0 <= rowb[0] + 15;
+    // NOTE: an `if (act0)`-guarded MFMA accumulation here (skipping the
+    // lower 16-row group on its fully-masked diagonal tiles) miscompiled —
+    // sparse wrong dqacc[0][0] values on waves 2/3 (hipcc/ROCm 7.2, wave-
+    // uniform condition derived from wid). Both groups are computed
+    // unconditionally; the causal mask already zeroes dS of masked tiles.
     {
       // ---- per mt: S^T = mfma(K_perm, Q), dP^T = mfma(V_perm, dO), then
       // immediately exp/pack dS into the dQ MFMA's A operand (keys in
       // kc*32+lg*8+j order via perm16) — keeps only one mt of S/dP live ----
       const bool smask = (kv_end < kv0 + 64);
-      uint32_t pk_ds[2][2][4];  // [nq][kc][4]
+      u32x4 pk_ds[2][2];  // [nq][kc]
+      const short* kl = k_lds[buf];
+      const short* vl = v_lds[buf];
 #pragma unroll
       for (int mt = 0; mt < 4; ++mt) {
-        int keyg = kv0 + perm16(mt, l15);
-        if (keyg >= S) keyg = S - 1;
-        const short* kp = kb + (int64_t)keyg * strideS_kv + lg * 8;
-        const short* vp = vb + (int64_t)keyg * strideS_kv + lg * 8;
+        const int kr = perm16(mt, l15);  // local key row
         f32x4 s0 = {0.f, 0.f, 0.f, 0.f}, s1 = {0.f, 0.f, 0.f, 0.f};
         f32x4 d0 = {0.f, 0.f, 0.f, 0.f}, d1 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int c = 0; c < 4; ++c)
           if (c < nkc) {
-            bf16x8 ka = *reinterpret_cast<const bf16x8*>(kp + c * 32);
-            bf16x8 va = *reinterpret_cast<const bf16x8*>(vp + c * 32);
-            if (act0) {
-              s0 = mfma16b(ka, qf[0][c], s0);
-              d0 = mfma16b(va, dof[0][c], d0);
-            }
+            bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+                kl + rm_idx(kr, c * 32 + lg * 8, D));
+            bf16x8 va = *reinterpret_cast<const bf16x8*>(
+                vl + rm_idx(kr, c * 32 + lg * 8, D));
+            s0 = mfma16b(ka, qf[0][c], s0);
+            d0 = mfma16b(va, dof[0][c], d0);
             s1 = mfma16b(ka, qf[1][c], s1);
             d1 = mfma16b(va, dof[1][c], d1);
           }
@@ -425,7 +417,6 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
         const int kc = mt >> 1, rp = (mt & 1) * 2;
 #pragma unroll
         for (int nq = 0; nq < 2; ++nq) {
-          if (nq == 0 && !act0) continue;
           const int qrow = rowb[nq] + l15;
           const bool diag = (kv0 + 63 > rowb[nq]) || smask;
           const f32x4 sv = nq ? s1 : s0;
@@ -445,24 +436,25 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
         }
       }
 
-      // ---- dQ += dS K (B = K^T-staged rows back in [key][d]... B operand
-      // [k=key][n=d] read from kt_lds transposed image) ----
-      const short* ktl = kt_lds[buf];
+      // ---- dQ += dS K: B operand [k=key][n=d] via tr16 transpose reads
+      // from the row-major K tile ----
 #pragma unroll
       for (int dt = 0; dt < 8; ++dt) {
         if (dt >= nd16) break;
-        const int d = dt * 16 + l15;
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
-          bf16x8 kfr = *reinterpret_cast<const bf16x8*>(
-              ktl + tr_idx(d, kc * 32 + lg * 8));
-          if (act0)
-            dqacc[dt][0] = mfma16b(
-                *reinterpret_cast<const bf16x8*>(&pk_ds[0][kc][0]), kfr,
-                dqacc[dt][0]);
-          dqacc[dt][1] = mfma16b(
-              *reinterpret_cast<const bf16x8*>(&pk_ds[1][kc][0]), kfr,
-              dqacc[dt][1]);
+#ifdef DQ_SCALAR_B
+          bf16x8 kfr;
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            kfr[j] = kl[rm_idx(kc * 32 + lg * 8 + j, dt * 16 + l15, D)];
+#else
+          bf16x8 kfr = tr16_frag(kl, kc * 32 + lg * 8, dt * 16, D, l15);
+#endif
+          dqacc[dt][0] = mfma16b(__builtin_bit_cast(bf16x8, pk_ds[0][kc]),
+                                 kfr, dqacc[dt][0]);
+          dqacc[dt][1] = mfma16b(__builtin_bit_cast(bf16x8, pk_ds[1][kc]),
+                                 kfr, dqacc[dt][1]);
         }
       }
     }

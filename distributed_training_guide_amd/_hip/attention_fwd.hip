@@ -51,20 +51,14 @@ __device__ __forceinline__ f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
 #define KVBLK 64   // keys per LDS tile
 #define LOG2E 1.4426950408889634f
 
-// v^T LDS image: [D up to 128][KVBLK] bf16, key index swizzled by d so the
-// transposed staging writes spread banks while the per-d 8-key block read
-// stays 16B-contiguous.
-__device__ __forceinline__ int vt_idx(int d, int key) {
-  return d * KVBLK + (key ^ ((d & 7) << 3) ^ (((d >> 3) & 7) << 3));
-}
-
 __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
     float* __restrict__ lse_out, int B, int S, int Hq, int Hkv, int D,
     float scale) {
-  __shared__ short vt_lds[2][128 * KVBLK];
-  __shared__ short k_lds[2][KVBLK * 128];  // row-major [key][d], slot-swizzled
+  // both tiles row-major [key][d], rm_idx slot-swizzled, double-buffered
+  __shared__ short k_lds[2][KVBLK * 128];
+  __shared__ short v_lds[2][KVBLK * 128];
 
   const int ntq = (S + QBLK - 1) / QBLK;
   const int gqa = Hq / Hkv;
@@ -153,33 +147,12 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
   const float c = scale * LOG2E;
   (void)hkv;
 
-  // ---- staging ----
-  // V: thread stages V rows transposed (key = tid&63, 16B of d per slot).
-  // K: staged ROW-major [key][d] with the 16B-slot XOR swizzle (16 lanes
-  //    cover one key row -> conflict-free b128 writes, coalesced reads);
-  //    both tiles are loaded ONE TILE AHEAD into the other buffer so the
-  //    global latency lands behind a full tile of MFMA (T14).
-  const int skey = tid & 63;
-  const int sslot0 = (tid >> 6) * 2;  // wave w covers d slots {2w,2w+1,8+2w,8+2w+1}
+  // ---- staging: K and V row-major [key][d], rm_idx-swizzled, 16 lanes
+  // per key row (conflict-free vectorized b128 writes, coalesced global
+  // reads); loaded ONE TILE AHEAD so the global latency lands behind a
+  // full tile of MFMA (T14) ----
   const int nslot = D >> 3;
   auto stage_kv = [&](int kv0, int buf) {
-    int keyg = kv0 + skey;
-    if (keyg >= S) keyg = S - 1;
-    const short* vp = vb + (int64_t)keyg * strideS_kv;
-    short* dst = vt_lds[buf];
-#pragma unroll
-    for (int half = 0; half < 2; ++half) {
-#pragma unroll
-      for (int ss = 0; ss < 2; ++ss) {
-        int slot = half * 8 + sslot0 + ss;  // d0 = slot*8
-        if (slot * 8 >= D) break;
-        bf16x8 vec = *reinterpret_cast<const bf16x8*>(vp + slot * 8);
-        int d0 = slot * 8;
-#pragma unroll
-        for (int i = 0; i < 8; ++i) dst[vt_idx(d0 + i, skey)] = vec[i];
-      }
-    }
-    short* kdst = k_lds[buf];
 #pragma unroll
     for (int vv = 0; vv < 4; ++vv) {
       int vecid = vv * 256 + tid;
@@ -187,10 +160,12 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
       if (slot >= nslot || key >= KVBLK) continue;
       int kg = kv0 + key;
       if (kg >= S) kg = S - 1;
-      bf16x8 vec = *reinterpret_cast<const bf16x8*>(
+      bf16x8 kv_ = *reinterpret_cast<const bf16x8*>(
           kb + (int64_t)kg * strideS_kv + slot * 8);
-      *reinterpret_cast<bf16x8*>(
-          kdst + key * D + ((slot ^ (key & 7)) & (nslot - 1)) * 8) = vec;
+      bf16x8 vv_ = *reinterpret_cast<const bf16x8*>(
+          vb + (int64_t)kg * strideS_kv + slot * 8);
+      *reinterpret_cast<bf16x8*>(k_lds[buf] + rm_idx(key, slot * 8, D)) = kv_;
+      *reinterpret_cast<bf16x8*>(v_lds[buf] + rm_idx(key, slot * 8, D)) = vv_;
     }
   };
 
@@ -204,10 +179,12 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
     // MFMA stream — T14 spirit; ds_writes don't touch the compute buffer)
     if (t + 1 < ntiles) stage_kv((t + 1) * KVBLK, buf ^ 1);
 
-    // per-16-row-group activity (wave-uniform): group 1 (upper rows) is
-    // live for every tile of this block; group 0 drops out only at the
-    // final diagonal tiles — uniform across waves, so no barrier idling
-    const bool act0 = kv0 <= rowb[0] + 15;
+    // Both 16-row groups are computed unconditionally on every tile: an
+    // `if (act0)`-guarded MFMA accumulation (skipping group 0 on its
+    // fully-masked diagonal tiles) miscompiled in the dq kernel of
+    // attention_bwd.hip (sparse wrong accumulator values; hipcc/ROCm 7.2)
+    // — the causal mask yields p=0 / alpha=1 there, so skipping is a
+    // ~3% optimization not worth the hazard.
 
     // ---- S^T = mfma(K, Q): [mt 4][nq 2] C frags; K A-frags from the
     // row-major staged tile (ds_read_b128, ~50cyc, hidden by MFMA) ----
@@ -221,8 +198,8 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
       for (int kc = 0; kc < 4; ++kc)
         if (kc < nkc) {
           bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-              kl + kr * D + (((kc * 4 + lg) ^ (kr & 7)) & (nslot - 1)) * 8);
-          if (act0) acc0 = mfma16(kf, qf[0][kc], acc0);
+              kl + rm_idx(kr, kc * 32 + lg * 8, D));
+          acc0 = mfma16(kf, qf[0][kc], acc0);
           acc1 = mfma16(kf, qf[1][kc], acc1);
         }
       sfrag[mt][0] = acc0;
@@ -230,11 +207,10 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
     }
 
     // ---- causal mask + online softmax (rows lane-local) ----
-    uint32_t pk[2][2][4];  // [nq][kc][4 regs of 2 bf16]
+    u32x4 pk[2][2];  // [nq][kc]: 4 regs of 2 bf16 (keys j=0..7)
     float alpha[2];
 #pragma unroll
     for (int nq = 0; nq < 2; ++nq) {
-      if (nq == 0 && !act0) continue;
       const int qrow = rowb[nq] + l15;
       const bool need_mask =
           (kv0 + KVBLK - 1) > rowb[nq] || kv_end < kv0 + KVBLK;
@@ -280,27 +256,23 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
       lrow[nq] = lrow[nq] * alpha[nq] + psum;
     }
 
-    // ---- O^T += mfma(V^T, P^T) ----
-    const short* vt = vt_lds[buf];
+    // ---- O^T += mfma(V^T, P^T): V^T A-frags via tr16 hardware-transpose
+    // reads from the row-major V tile (no transposed copy) ----
+    const short* vt = v_lds[buf];
 #pragma unroll
     for (int dt = 0; dt < 8; ++dt) {
       if (dt >= nd16) break;
-      const int d = dt * 16 + l15;
       bf16x8 va[2];
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc)
-        va[kc] = *reinterpret_cast<const bf16x8*>(
-            vt + vt_idx(d, kc * 32 + lg * 8));
+        va[kc] = tr16_frag(vt, kc * 32 + lg * 8, dt * 16, D, l15);
 #pragma unroll
       for (int nq = 0; nq < 2; ++nq) {
-        if (nq == 0 && !act0) continue;
         f32x4 acc = oacc[dt][nq];
 #pragma unroll
         for (int r = 0; r < 4; ++r) acc[r] *= alpha[nq];
-        acc = mfma16(va[0], *reinterpret_cast<const bf16x8*>(&pk[nq][0][0]),
-                     acc);
-        acc = mfma16(va[1], *reinterpret_cast<const bf16x8*>(&pk[nq][1][0]),
-                     acc);
+        acc = mfma16(va[0], __builtin_bit_cast(bf16x8, pk[nq][0]), acc);
+        acc = mfma16(va[1], __builtin_bit_cast(bf16x8, pk[nq][1]), acc);
         oacc[dt][nq] = acc;
       }
     }

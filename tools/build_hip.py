@@ -43,6 +43,8 @@ def _common_flags(includes):
         "-DTORCH_API_INCLUDE_EXTENSION_H",
         "-DTORCH_EXTENSION_NAME=_C",
     ]
+    if os.environ.get("DTGA_HIP_FLAGS"):
+        flags += os.environ["DTGA_HIP_FLAGS"].split()
     for inc in includes:
         flags.append(f"-I{inc}")
     flags.append(f"-I{sysconfig.get_paths()['include']}")
