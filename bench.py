@@ -37,8 +37,10 @@ def parse_args():
     p.add_argument("--batch-size", type=int, default=16, help="per GPU")
     p.add_argument("--seq-length", type=int, default=1024)
     p.add_argument("--bucket-cap-mb", type=int, default=128)
-    p.add_argument("--no-zero1", dest="zero1", action="store_false",
-                   help="ZeRO-1 optimizer sharding (on by default for N>1)")
+    p.add_argument("--zero1", action="store_true",
+                   help="ZeRO-1 optimizer sharding (off by default: on one "
+                        "288GB-HBM node the replicated fused AdamW avoids "
+                        "the un-overlapped post-step shard all-gather)")
     p.add_argument("--device", default=None)
     return p.parse_args()
 
