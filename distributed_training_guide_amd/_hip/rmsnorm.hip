@@ -53,63 +53,95 @@ __global__ void __launch_bounds__(256) rmsnorm_fwd_kernel(
 }
 
 // ---------------- backward: dx + per-block dw partials ----------------
+// dw partials accumulate in REGISTERS across this block's rows (each
+// thread owns fixed columns) and spill to dw_partial ONCE at the end;
+// dy/x rows are staged in dynamic LDS on pass 1 so pass 2 never re-reads
+// global. Traffic/row: read dy+x, write dx (the memory-bound floor).
 __global__ void __launch_bounds__(256) rmsnorm_bwd_kernel(
     const short* __restrict__ dy, const short* __restrict__ x,
     const short* __restrict__ w, const float* __restrict__ rstd,
     short* __restrict__ dx, float* __restrict__ dw_partial,
     int64_t nrows, int H) {
   __shared__ float scratch[8];
+  extern __shared__ short rowbuf[];  // [H] dy then [H] x
+  short* dy_l = rowbuf;
+  short* x_l = rowbuf + H;
   const float invH = 1.0f / (float)H;
-  float* dwp = dw_partial + (int64_t)blockIdx.x * H;
-  // zero this block's partial row
-  for (int i = threadIdx.x; i < H; i += blockDim.x) dwp[i] = 0.0f;
-  __syncthreads();
+  const bool vec = (H & 7) == 0 && H <= 2048 * 8;
+  // per-thread dw accumulators: chunk c covers column i = c*2048 + tid*8
+  const int nch = (H + 2047) / 2048;
+  float dwacc[8][8];
+#pragma unroll
+  for (int c = 0; c < 8; ++c)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dwacc[c][j] = 0.0f;
+
   for (int64_t row = blockIdx.x; row < nrows; row += gridDim.x) {
     const short* dyr = dy + row * H;
     const short* xr = x + row * H;
     short* dxr = dx + row * H;
     const float rs = rstd[row];
-    // pass 1: dot = mean(g * xhat)
     float dot = 0.0f;
-    if ((H & 7) == 0) {
+    if (vec) {
       for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
         s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i);
         s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i);
         s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+        *reinterpret_cast<s16x8*>(dy_l + i) = dv;
+        *reinterpret_cast<s16x8*>(x_l + i) = xv;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           dot += bf2f(dv[j]) * bf2f(wv[j]) * bf2f(xv[j]) * rs;
       }
     } else {
-      for (int i = threadIdx.x; i < H; i += blockDim.x)
+      for (int i = threadIdx.x; i < H; i += blockDim.x) {
+        dy_l[i] = dyr[i];
+        x_l[i] = xr[i];
         dot += bf2f(dyr[i]) * bf2f(w[i]) * bf2f(xr[i]) * rs;
+      }
     }
     dot = block_reduce_sum(dot, scratch) * invH;
-    // pass 2: dx and dw partial accumulate
-    if ((H & 7) == 0) {
-      for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
-        s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i);
-        s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i);
+    if (vec) {
+      int c = 0;
+      for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8, ++c) {
+        s16x8 dv = *reinterpret_cast<const s16x8*>(dy_l + i);
+        s16x8 xv = *reinterpret_cast<const s16x8*>(x_l + i);
         s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
         s16x8 o;
+        float* acc = dwacc[c];
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float g = bf2f(dv[j]) * bf2f(wv[j]);
           float xhat = bf2f(xv[j]) * rs;
-          o[j] = f2bf(rs * (g - xhat * dot));  // dx = rstd*(dy*w - xhat*mean(g*xhat))
-          dwp[i + j] += bf2f(dv[j]) * xhat;
+          o[j] = f2bf(rs * (g - xhat * dot));
+          acc[j] += bf2f(dv[j]) * xhat;
         }
         *reinterpret_cast<s16x8*>(dxr + i) = o;
       }
     } else {
+      // scalar fallback: accumulate straight to the partial row (rare)
+      float* dwp = dw_partial + (int64_t)blockIdx.x * H;
       for (int i = threadIdx.x; i < H; i += blockDim.x) {
-        float g = bf2f(dyr[i]) * bf2f(w[i]);
-        float xhat = bf2f(xr[i]) * rs;
+        float g = bf2f(dy_l[i]) * bf2f(w[i]);
+        float xhat = bf2f(x_l[i]) * rs;
         dxr[i] = f2bf(rs * (g - xhat * dot));
-        dwp[i] += bf2f(dyr[i]) * xhat;
+        dwp[i] += bf2f(dy_l[i]) * xhat;
       }
     }
     __syncthreads();
+  }
+  // spill register accumulators once
+  float* dwp = dw_partial + (int64_t)blockIdx.x * H;
+  if (vec) {
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      if (c >= nch) break;
+      int i = c * 2048 + threadIdx.x * 8;
+      if (i < H) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) dwp[i + j] = dwacc[c][j];
+      }
+    }
   }
 }
 
@@ -146,7 +178,12 @@ void rmsnorm_bwd_launch(const void* dy, const void* x, const void* w,
                         const void* rstd, void* dx, float* dw_partial,
                         void* dw, int nblocks, int64_t nrows, int H,
                         hipStream_t s) {
-  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(nblocks), dim3(256), 0, s,
+  // scalar fallback (H not a multiple of 8) accumulates into dw_partial
+  // directly, so it must start zeroed; the vectorized path overwrites.
+  if ((H & 7) || H > 2048 * 8)
+    hipMemsetAsync(dw_partial, 0, (size_t)nblocks * H * sizeof(float), s);
+  size_t shmem = 2 * (size_t)H * sizeof(short);
+  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(nblocks), dim3(256), shmem, s,
                      (const short*)dy, (const short*)x, (const short*)w,
                      (const float*)rstd, (short*)dx, dw_partial, nrows, H);
   int rgrid = (H + 255) / 256;
