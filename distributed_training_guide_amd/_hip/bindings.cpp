@@ -62,7 +62,7 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   CHECK_BF16_CONTIG(x);
   const int H = (int)x.size(-1);
   const int64_t nrows = x.numel() / H;
-  const int nblocks = (int)std::min<int64_t>(nrows, 256);
+  const int nblocks = (int)std::min<int64_t>(nrows, 2048);
   auto dx = torch::empty_like(x);
   auto dw = torch::empty_like(w);
   auto dw_partial =

@@ -88,6 +88,7 @@ def test_silu_mul_fwd_bwd():
     dict(B=1, S=1024, Hq=4, Hkv=1, D=128),
     dict(B=1, S=200, Hq=2, Hkv=2, D=128),  # ragged S
     dict(B=1, S=48, Hq=2, Hkv=2, D=64),    # S < tile
+    dict(B=1, S=4096, Hq=4, Hkv=1, D=128),  # chapter-5 seq length
 ])
 def test_attention_fwd_bwd(cfg):
     torch.manual_seed(0)
