@@ -148,25 +148,36 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
   // conflict-free), one (h, q-tile) AHEAD so global latency hides behind
   // a full tile of MFMA
   const int nslot = D >> 3;
-  auto stage = [&](int it, int buf) {
+  const int srow0 = tid / 16;  // thread stages rows srow0 + {0,16,32,48}
+  const int sslot = tid & 15;
+  const bool svalid = sslot < nslot;
+  // T14 split staging: issue the next (h, q-tile)'s global loads into
+  // registers early (latency hides under this tile's MFMA stream), write
+  // them to LDS just before the barrier.
+  auto issue2 = [&](int it, const short* base0, bf16x8* out) {
     const int hh = hkv * group + it / nqt;
     const int qt = qstart + (it % nqt) * 64;
-    const short* qbase = q + ((int64_t)b * S * Hq + hh) * D;
-    const short* dobase = dout + ((int64_t)b * S * Hq + hh) * D;
+    const short* bse = base0 + ((int64_t)b * S * Hq + hh) * D;
+    if (svalid) {
 #pragma unroll
-    for (int vv = 0; vv < 4; ++vv) {
-      int vecid = vv * 256 + tid;
-      int row = vecid / 16, slot = vecid & 15;
-      if (slot >= nslot || row >= 64) continue;
-      int qrow = qt + row;
-      if (qrow >= S) qrow = S - 1;
-      bf16x8 qv = *reinterpret_cast<const bf16x8*>(
-          qbase + (int64_t)qrow * strideS_q + slot * 8);
-      bf16x8 dv = *reinterpret_cast<const bf16x8*>(
-          dobase + (int64_t)qrow * strideS_q + slot * 8);
-      *reinterpret_cast<bf16x8*>(q_lds[buf] + rm_idx(row, slot * 8, D)) = qv;
-      *reinterpret_cast<bf16x8*>(do_lds[buf] + rm_idx(row, slot * 8, D)) = dv;
+      for (int vv = 0; vv < 4; ++vv) {
+        int r = qt + srow0 + vv * 16;
+        if (r >= S) r = S - 1;
+        out[vv] = *reinterpret_cast<const bf16x8*>(
+            bse + (int64_t)r * strideS_q + sslot * 8);
+      }
     }
+  };
+  auto write2 = [&](short* dst, const bf16x8* vecs) {
+    if (!svalid) return;
+#pragma unroll
+    for (int vv = 0; vv < 4; ++vv)
+      *reinterpret_cast<bf16x8*>(
+          dst + rm_idx(srow0 + vv * 16, sslot * 8, D)) = vecs[vv];
+  };
+  auto stage_lse = [&](int it, int buf) {
+    const int hh = hkv * group + it / nqt;
+    const int qt = qstart + (it % nqt) * 64;
     if (tid < 64) {
       const float* lseb = lse + ((int64_t)b * Hq + hh) * S;
       int rr = qt + tid < S ? qt + tid : S - 1;
@@ -178,20 +189,36 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
     }
   };
 
-  stage(0, 0);
+  {
+    bf16x8 qv[4], dv[4];
+    issue2(0, q, qv);
+    issue2(0, dout, dv);
+    write2(q_lds[0], qv);
+    write2(do_lds[0], dv);
+    stage_lse(0, 0);
+  }
   __syncthreads();
 
   for (int it = 0; it < niter; ++it) {
     const int buf = it & 1;
-    if (it + 1 < niter) stage(it + 1, buf ^ 1);
+    const bool more = it + 1 < niter;
+    bf16x8 qnext[4];
+    if (more) {
+      issue2(it + 1, q, qnext);
+      stage_lse(it + 1, buf ^ 1);
+    }
 
     const int qt = qstart + (it % nqt) * 64;
     const short* ql = q_lds[buf];
     const short* dol = do_lds[buf];
 
-    // ---- S[mt], dP[mt]: C[m=qrow(perm)][n=key]; A = Q/dO rows fed in
-    // perm16 order (b128 LDS reads), B = kf/vf ----
-    f32x4 sfrag[4], dpfrag[4];
+    // ---- per mt: S/dP MFMAs (C[m=qrow(perm)][n=key]; A = Q/dO rows in
+    // perm16 order from LDS, B = kf/vf), then immediately
+    // P = exp(scale*S - lse), dS = scale*P*(dP - delta), packed into
+    // A-operand order — only one mt of S/dP state stays live ----
+    const int key = kv0 + l15;  // this lane's key (C n-position)
+    const bool diag = (qt < kt * 64 + 64) || (qt + 63 >= S);
+    u32x4 pk_p[2], pk_ds[2];
 #pragma unroll
     for (int mt = 0; mt < 4; ++mt) {
       const int qr = perm16(mt, l15);  // local row in the staged tile
@@ -206,27 +233,16 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
           sa = mfma16b(qa, kf[c], sa);
           da = mfma16b(doa, vf[c], da);
         }
-      sfrag[mt] = sa;
-      dpfrag[mt] = da;
-    }
-
-    // ---- P = exp(scale*S - lse), dS = scale*P*(dP - delta); pack into
-    // A-operand order (qrow k-dim = kc*32+lg*8+j via perm16) ----
-    const int key = kv0 + l15;  // this lane's key (C n-position)
-    const bool diag = (qt < kt * 64 + 64) || (qt + 63 >= S);
-    u32x4 pk_p[2], pk_ds[2];
-#pragma unroll
-    for (int mt = 0; mt < 4; ++mt) {
       const int qoff = cpos16(mt, lg);
       float p[4], ds[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int qloc = qoff + r;
         const int qrow = qt + qloc;
-        float e = (scale * sfrag[mt][r] - lse_lds[buf][qloc]) * LOG2E;
+        float e = (scale * sa[r] - lse_lds[buf][qloc]) * LOG2E;
         if (diag && (key > qrow || qrow >= S || key >= S)) e = -INFINITY;
         p[r] = exp2f(e);
-        ds[r] = scale * p[r] * (dpfrag[mt][r] - del_lds[buf][qloc]);
+        ds[r] = scale * p[r] * (da[r] - del_lds[buf][qloc]);
       }
       const int kc = mt >> 1, rp = (mt & 1) * 2;
       pk_p[kc][rp + 0] = cvt_pk_bf16(p[0], p[1]);
@@ -234,6 +250,12 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
       pk_ds[kc][rp + 0] = cvt_pk_bf16(ds[0], ds[1]);
       pk_ds[kc][rp + 1] = cvt_pk_bf16(ds[2], ds[3]);
     }
+
+    // q write-back here, dO issue now: the two staging register windows
+    // never overlap (VGPR budget), and dO's latency hides under dV/dK
+    if (more) write2(q_lds[buf ^ 1], qnext);
+    bf16x8 donext[4];
+    if (more) issue2(it + 1, dout, donext);
 
     // ---- dV += P^T dO ; dK += dS^T Q (B operands via tr16 transpose
     // reads from the same row-major tiles) ----
@@ -250,6 +272,7 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
                             dkacc[dt]);
       }
     }
+    if (more) write2(do_lds[buf ^ 1], donext);
     __syncthreads();
   }
 
