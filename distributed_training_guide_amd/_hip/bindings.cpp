@@ -10,6 +10,12 @@ void rmsnorm_fwd_launch(const void*, const void*, void*, void*, int64_t, int,
                         float, hipStream_t);
 void rmsnorm_bwd_launch(const void*, const void*, const void*, const void*,
                         void*, float*, void*, int, int64_t, int, hipStream_t);
+void qkv_rope_fwd_launch(const void*, void*, void*, void*, const float*,
+                         const float*, const int*, int64_t, int, int, int,
+                         int, hipStream_t);
+void qkv_rope_bwd_launch(const void*, const void*, const void*, void*,
+                         const float*, const float*, const int*, int64_t,
+                         int, int, int, int, hipStream_t);
 void rope_launch(const void*, void*, const float*, const float*, const int*,
                  int64_t, int, int, int, int, hipStream_t);
 void silu_mul_fwd_launch(const void*, void*, int64_t, int, hipStream_t);
@@ -95,6 +101,52 @@ torch::Tensor rope(torch::Tensor x, torch::Tensor cos, torch::Tensor sin,
               sin.data_ptr<float>(), pos_ptr, x.numel() / D, S, H, D,
               backward ? 1 : 0, cur_stream());
   return y;
+}
+
+// ---------------- fused qkv split + rope ----------------
+std::vector<torch::Tensor> qkv_rope_fwd(torch::Tensor qkv, torch::Tensor cos,
+                                        torch::Tensor sin,
+                                        c10::optional<torch::Tensor> positions,
+                                        int64_t Hq, int64_t Hkv, int64_t D) {
+  CHECK_BF16_CONTIG(qkv);
+  TORCH_CHECK(qkv.dim() == 3, "qkv_rope expects [B,S,W]");
+  const int64_t B = qkv.size(0), S = qkv.size(1);
+  TORCH_CHECK(qkv.size(2) == (Hq + 2 * Hkv) * D, "packed width mismatch");
+  const int* pos_ptr = nullptr;
+  if (positions.has_value()) {
+    TORCH_CHECK(positions->dtype() == torch::kInt &&
+                positions->is_contiguous() && positions->numel() == S);
+    pos_ptr = positions->data_ptr<int>();
+  } else {
+    TORCH_CHECK(cos.size(0) >= S, "rope table shorter than sequence");
+  }
+  auto q = torch::empty({B, S, Hq, D}, qkv.options());
+  auto k = torch::empty({B, S, Hkv, D}, qkv.options());
+  auto v = torch::empty({B, S, Hkv, D}, qkv.options());
+  qkv_rope_fwd_launch(qkv.data_ptr(), q.data_ptr(), k.data_ptr(),
+                      v.data_ptr(), cos.data_ptr<float>(),
+                      sin.data_ptr<float>(), pos_ptr, B * S, (int)S, (int)Hq,
+                      (int)Hkv, (int)D, cur_stream());
+  return {q, k, v};
+}
+
+torch::Tensor qkv_rope_bwd(torch::Tensor dq, torch::Tensor dk,
+                           torch::Tensor dv, torch::Tensor cos,
+                           torch::Tensor sin,
+                           c10::optional<torch::Tensor> positions) {
+  CHECK_BF16_CONTIG(dq);
+  CHECK_BF16_CONTIG(dk);
+  CHECK_BF16_CONTIG(dv);
+  const int64_t B = dq.size(0), S = dq.size(1);
+  const int64_t Hq = dq.size(2), Hkv = dk.size(2), D = dq.size(3);
+  const int* pos_ptr = nullptr;
+  if (positions.has_value()) pos_ptr = positions->data_ptr<int>();
+  auto dqkv = torch::empty({B, S, (Hq + 2 * Hkv) * D}, dq.options());
+  qkv_rope_bwd_launch(dq.data_ptr(), dk.data_ptr(), dv.data_ptr(),
+                      dqkv.data_ptr(), cos.data_ptr<float>(),
+                      sin.data_ptr<float>(), pos_ptr, B * S, (int)S, (int)Hq,
+                      (int)Hkv, (int)D, cur_stream());
+  return dqkv;
 }
 
 // ---------------- silu_mul ----------------
@@ -234,6 +286,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("rope", &rope);
+  m.def("qkv_rope_fwd", &qkv_rope_fwd);
+  m.def("qkv_rope_bwd", &qkv_rope_bwd);
   m.def("silu_mul_fwd", &silu_mul_fwd);
   m.def("silu_mul_bwd", &silu_mul_bwd);
   m.def("ce_fwd", &ce_fwd);

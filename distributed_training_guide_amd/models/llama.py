@@ -21,7 +21,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops import RMSNorm, causal_lm_loss, flash_attention, rope, silu_mul
+from ..ops import (RMSNorm, causal_lm_loss, flash_attention, qkv_rope,
+                   silu_mul)
 
 
 def pad_vocab(v: int, multiple: int = 64) -> int:
@@ -87,15 +88,11 @@ class LlamaAttention(nn.Module):
         B, S, _ = x.shape
         d = self.head_dim
         qkv = self.qkv_proj(x)
-        q, k, v = qkv.split([self.num_heads * d, self.num_kv_heads * d,
-                             self.num_kv_heads * d], dim=-1)
-        q = q.view(B, S, self.num_heads, d)
-        k = k.view(B, S, self.num_kv_heads, d)
-        v = v.view(B, S, self.num_kv_heads, d).contiguous()
-        theta = self.config.rope_theta
-        maxp = self.config.max_position_embeddings
-        q = rope(q, theta, positions=position_ids, max_pos=maxp)
-        k = rope(k, theta, positions=position_ids, max_pos=maxp)
+        # fused split + RoPE: one pass, no intermediate copies, backward
+        # writes dqkv directly (no grad cat)
+        q, k, v = qkv_rope(qkv, self.num_heads, self.num_kv_heads, d,
+                           self.config.rope_theta, positions=position_ids,
+                           max_pos=self.config.max_position_embeddings)
         o = flash_attention(q, k, v)
         return self.o_proj(o.reshape(B, S, self.num_heads * d))
 

@@ -3,7 +3,7 @@ from .adamw import FusedAdamW
 from .attention import flash_attention
 from .cross_entropy import causal_lm_loss, sharded_causal_lm_loss
 from .rmsnorm import RMSNorm, rmsnorm
-from .rope import rope
+from .rope import qkv_rope, rope
 from .swiglu import silu_mul
 
 __all__ = [
@@ -13,6 +13,7 @@ __all__ = [
     "sharded_causal_lm_loss",
     "RMSNorm",
     "rmsnorm",
+    "qkv_rope",
     "rope",
     "silu_mul",
 ]

@@ -37,6 +37,48 @@ class _RoPEFn(torch.autograd.Function):
         return dx, None, None, None
 
 
+class _QKVRopeFn(torch.autograd.Function):
+    """Fused qkv split + RoPE (qkv_rope.hip): one pass produces rotated
+    contiguous q/k plus v from the packed projection output; backward
+    re-packs dq/dk/dv into dqkv with the inverse rotation (replacing the
+    split -> rope -> contiguous chain and the grad cat)."""
+
+    @staticmethod
+    def forward(ctx, qkv, cos, sin, positions, Hq, Hkv, D):
+        q, k, v = ext().qkv_rope_fwd(qkv, cos, sin, positions, Hq, Hkv, D)
+        ctx.cos, ctx.sin, ctx.positions = cos, sin, positions
+        return q, k, v
+
+    @staticmethod
+    def backward(ctx, dq, dk, dv):
+        dqkv = ext().qkv_rope_bwd(dq.contiguous(), dk.contiguous(),
+                                  dv.contiguous(), ctx.cos, ctx.sin,
+                                  ctx.positions)
+        return dqkv, None, None, None, None, None, None
+
+
+def qkv_rope(qkv: torch.Tensor, Hq: int, Hkv: int, D: int,
+             theta: float = 10000.0,
+             positions: torch.Tensor | None = None,
+             max_pos: int | None = None):
+    """Split packed qkv [B,S,(Hq+2Hkv)*D] and apply RoPE to q/k in one
+    fused pass. Returns (q [B,S,Hq,D], k, v [B,S,Hkv,D])."""
+    B, S, W = qkv.shape
+    need = max_pos or S
+    if positions is not None:
+        positions = positions.to(torch.int32).contiguous()
+        need = max(need, int(positions.max().item()) + 1)
+    cos, sin = get_rope_tables(D, need, theta, qkv.device)
+    if use_hip(qkv):
+        return _QKVRopeFn.apply(qkv.contiguous(), cos, sin, positions,
+                                Hq, Hkv, D)
+    # CPU fallback: split views + eager rope
+    q, k, v = qkv.split([Hq * D, Hkv * D, Hkv * D], dim=-1)
+    q = rope_ref(q.view(B, S, Hq, D), cos, sin, positions)
+    k = rope_ref(k.view(B, S, Hkv, D), cos, sin, positions)
+    return q, k, v.view(B, S, Hkv, D).contiguous()
+
+
 def rope(x: torch.Tensor, theta: float = 10000.0,
          positions: torch.Tensor | None = None,
          max_pos: int | None = None) -> torch.Tensor:

@@ -185,3 +185,29 @@ def test_model_train_step_gpu():
         losses.append(out.loss.item())
     assert all(math.isfinite(l) for l in losses)
     assert losses[-1] < losses[0]  # memorizing one batch must reduce loss
+
+
+def test_qkv_rope_fused_matches_unfused():
+    from distributed_training_guide_amd.ops import qkv_rope, rope
+
+    torch.manual_seed(3)
+    B, S, Hq, Hkv, D = 2, 256, 8, 2, 64
+    W = (Hq + 2 * Hkv) * D
+    qkv = torch.randn(B, S, W, device="cuda", dtype=torch.bfloat16,
+                      requires_grad=True)
+    q, k, v = qkv_rope(qkv, Hq, Hkv, D, theta=1e4)
+    # unfused reference path
+    qkv2 = qkv.detach().clone().requires_grad_(True)
+    q2, k2, v2 = qkv2.split([Hq * D, Hkv * D, Hkv * D], dim=-1)
+    q2 = rope(q2.view(B, S, Hq, D), 1e4)
+    k2 = rope(k2.view(B, S, Hkv, D), 1e4)
+    v2 = v2.view(B, S, Hkv, D).contiguous()
+    assert torch.equal(v, v2)
+    assert _rel_err(q, q2.float()) < 1e-2
+    assert _rel_err(k, k2.float()) < 1e-2
+    dq = torch.randn_like(q)
+    dk = torch.randn_like(k)
+    dv = torch.randn_like(v)
+    torch.autograd.backward([q, k, v], [dq, dk, dv])
+    torch.autograd.backward([q2, k2, v2], [dq, dk, dv])
+    assert _rel_err(qkv.grad, qkv2.grad.float()) < 1e-2
