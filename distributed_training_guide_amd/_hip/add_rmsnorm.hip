@@ -1,0 +1,146 @@
+// Fused residual-add + RMSNorm for gfx950.
+//
+// The decoder stack threads a (delta, residual) pair between layers:
+//   res_out = residual + delta;  y = rmsnorm(res_out) * w
+// fused into one pass (read res+delta, write res_out+y) — the unfused
+// chain costs an extra full read+write of the hidden state per boundary
+// plus the backward's gradient-accumulation elementwise adds.
+//
+// Backward (given dy = d_normed and dr = d_res_out):
+//   g = dy * w, xhat = res_out * rstd
+//   dx = rstd * (g - xhat * mean(g * xhat)) + dr      (d_residual = d_delta)
+//   dw += dy * xhat   (register partials, one spill per block — same
+//                      scheme as rmsnorm.hip)
+#include "common.h"
+
+__global__ void __launch_bounds__(256) add_rmsnorm_fwd_kernel(
+    const short* __restrict__ res, const short* __restrict__ delta,
+    const short* __restrict__ w, short* __restrict__ res_out,
+    short* __restrict__ y, float* __restrict__ rstd_out, int64_t nrows,
+    int H, float eps) {
+  __shared__ float scratch[8];
+  extern __shared__ short rowbuf[];  // [H] staged res_out (bf16)
+  const float invH = 1.0f / (float)H;
+  for (int64_t row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const short* rr = res + row * H;
+    const short* dr = delta + row * H;
+    short* ro = res_out + row * H;
+    short* yr = y + row * H;
+    float ss = 0.0f;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+      s16x8 rv = *reinterpret_cast<const s16x8*>(rr + i);
+      s16x8 dv = *reinterpret_cast<const s16x8*>(dr + i);
+      s16x8 sv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(rv[j]) + bf2f(dv[j]);
+        sv[j] = f2bf(f);
+        float fq = bf2f(sv[j]);  // accumulate on the ROUNDED value so the
+        ss += fq * fq;           // norm matches what res_out stores
+      }
+      *reinterpret_cast<s16x8*>(ro + i) = sv;
+      *reinterpret_cast<s16x8*>(rowbuf + i) = sv;
+    }
+    ss = block_reduce_sum(ss, scratch);
+    const float rstd = rsqrtf(ss * invH + eps);
+    if (threadIdx.x == 0 && rstd_out != nullptr) rstd_out[row] = rstd;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+      s16x8 sv = *reinterpret_cast<const s16x8*>(rowbuf + i);
+      s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = f2bf(bf2f(sv[j]) * rstd * bf2f(wv[j]));
+      *reinterpret_cast<s16x8*>(yr + i) = o;
+    }
+    __syncthreads();
+  }
+}
+
+__global__ void __launch_bounds__(256) add_rmsnorm_bwd_kernel(
+    const short* __restrict__ dy, const short* __restrict__ dres_out,
+    const short* __restrict__ x /* = res_out */, const short* __restrict__ w,
+    const float* __restrict__ rstd, short* __restrict__ dx,
+    float* __restrict__ dw_partial, int64_t nrows, int H) {
+  __shared__ float scratch[8];
+  extern __shared__ short rowbuf[];  // [H] dy then [H] x
+  short* dy_l = rowbuf;
+  short* x_l = rowbuf + H;
+  const float invH = 1.0f / (float)H;
+  const int nch = (H + 2047) / 2048;
+  float dwacc[8][8];
+#pragma unroll
+  for (int c = 0; c < 8; ++c)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dwacc[c][j] = 0.0f;
+
+  for (int64_t row = blockIdx.x; row < nrows; row += gridDim.x) {
+    const short* dyr = dy + row * H;
+    const short* xr = x + row * H;
+    const short* drr = dres_out + row * H;
+    short* dxr = dx + row * H;
+    const float rs = rstd[row];
+    float dot = 0.0f;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+      s16x8 dv = *reinterpret_cast<const s16x8*>(dyr + i);
+      s16x8 xv = *reinterpret_cast<const s16x8*>(xr + i);
+      s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+      *reinterpret_cast<s16x8*>(dy_l + i) = dv;
+      *reinterpret_cast<s16x8*>(x_l + i) = xv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dot += bf2f(dv[j]) * bf2f(wv[j]) * bf2f(xv[j]) * rs;
+    }
+    dot = block_reduce_sum(dot, scratch) * invH;
+    int c = 0;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8, ++c) {
+      s16x8 dv = *reinterpret_cast<const s16x8*>(dy_l + i);
+      s16x8 xv = *reinterpret_cast<const s16x8*>(x_l + i);
+      s16x8 wv = *reinterpret_cast<const s16x8*>(w + i);
+      s16x8 rv = *reinterpret_cast<const s16x8*>(drr + i);
+      s16x8 o;
+      float* acc = dwacc[c];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = bf2f(dv[j]) * bf2f(wv[j]);
+        float xhat = bf2f(xv[j]) * rs;
+        o[j] = f2bf(rs * (g - xhat * dot) + bf2f(rv[j]));
+        acc[j] += bf2f(dv[j]) * xhat;
+      }
+      *reinterpret_cast<s16x8*>(dxr + i) = o;
+    }
+    __syncthreads();
+  }
+  float* dwp = dw_partial + (int64_t)blockIdx.x * H;
+#pragma unroll
+  for (int c = 0; c < 8; ++c) {
+    if (c >= nch) break;
+    int i = c * 2048 + threadIdx.x * 8;
+    if (i < H) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dwp[i + j] = dwacc[c][j];
+    }
+  }
+}
+
+extern "C" {
+void add_rmsnorm_fwd_launch(const void* res, const void* delta, const void* w,
+                            void* res_out, void* y, void* rstd, int64_t nrows,
+                            int H, float eps, hipStream_t s) {
+  int grid = (int)(nrows < 2048 ? (nrows < 1 ? 1 : nrows) : 2048);
+  size_t shmem = (size_t)H * sizeof(short);
+  hipLaunchKernelGGL(add_rmsnorm_fwd_kernel, dim3(grid), dim3(256), shmem, s,
+                     (const short*)res, (const short*)delta, (const short*)w,
+                     (short*)res_out, (short*)y, (float*)rstd, nrows, H, eps);
+}
+void add_rmsnorm_bwd_launch(const void* dy, const void* dres_out,
+                            const void* x, const void* w, const void* rstd,
+                            void* dx, float* dw_partial, int nblocks,
+                            int64_t nrows, int H, hipStream_t s) {
+  size_t shmem = 2 * (size_t)H * sizeof(short);
+  hipLaunchKernelGGL(add_rmsnorm_bwd_kernel, dim3(nblocks), dim3(256), shmem,
+                     s, (const short*)dy, (const short*)dres_out,
+                     (const short*)x, (const short*)w, (const float*)rstd,
+                     (short*)dx, dw_partial, nrows, H);
+}
+}

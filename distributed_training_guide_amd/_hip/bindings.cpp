@@ -16,6 +16,12 @@ void qkv_rope_fwd_launch(const void*, void*, void*, void*, const float*,
 void qkv_rope_bwd_launch(const void*, const void*, const void*, void*,
                          const float*, const float*, const int*, int64_t,
                          int, int, int, int, hipStream_t);
+void add_rmsnorm_fwd_launch(const void*, const void*, const void*, void*,
+                            void*, void*, int64_t, int, float, hipStream_t);
+void add_rmsnorm_bwd_launch(const void*, const void*, const void*,
+                            const void*, const void*, void*, float*, int,
+                            int64_t, int, hipStream_t);
+void rmsnorm_dw_reduce_launch(const float*, void*, int, int, hipStream_t);
 void rope_launch(const void*, void*, const float*, const float*, const int*,
                  int64_t, int, int, int, int, hipStream_t);
 void silu_mul_fwd_launch(const void*, void*, int64_t, int, hipStream_t);
@@ -101,6 +107,50 @@ torch::Tensor rope(torch::Tensor x, torch::Tensor cos, torch::Tensor sin,
               sin.data_ptr<float>(), pos_ptr, x.numel() / D, S, H, D,
               backward ? 1 : 0, cur_stream());
   return y;
+}
+
+// ---------------- fused residual add + rmsnorm ----------------
+std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor res,
+                                           torch::Tensor delta,
+                                           torch::Tensor w, double eps) {
+  CHECK_BF16_CONTIG(res);
+  CHECK_BF16_CONTIG(delta);
+  CHECK_BF16_CONTIG(w);
+  const int H = (int)res.size(-1);
+  TORCH_CHECK(H % 8 == 0 && H <= 16384, "add_rmsnorm: unsupported H");
+  const int64_t nrows = res.numel() / H;
+  auto res_out = torch::empty_like(res);
+  auto y = torch::empty_like(res);
+  auto rstd = torch::empty({nrows}, res.options().dtype(torch::kFloat));
+  add_rmsnorm_fwd_launch(res.data_ptr(), delta.data_ptr(), w.data_ptr(),
+                         res_out.data_ptr(), y.data_ptr(), rstd.data_ptr(),
+                         nrows, H, (float)eps, cur_stream());
+  return {y, res_out, rstd};
+}
+
+std::vector<torch::Tensor> add_rmsnorm_bwd(torch::Tensor dy,
+                                           torch::Tensor dres_out,
+                                           torch::Tensor res_out,
+                                           torch::Tensor w,
+                                           torch::Tensor rstd) {
+  CHECK_BF16_CONTIG(dy);
+  CHECK_BF16_CONTIG(dres_out);
+  CHECK_BF16_CONTIG(res_out);
+  const int H = (int)res_out.size(-1);
+  const int64_t nrows = res_out.numel() / H;
+  const int nblocks = (int)std::min<int64_t>(nrows, 2048);
+  auto dx = torch::empty_like(res_out);
+  auto dw = torch::empty_like(w);
+  auto dw_partial =
+      torch::empty({(int64_t)nblocks, (int64_t)H},
+                   res_out.options().dtype(torch::kFloat));
+  add_rmsnorm_bwd_launch(dy.data_ptr(), dres_out.data_ptr(),
+                         res_out.data_ptr(), w.data_ptr(), rstd.data_ptr(),
+                         dx.data_ptr(), dw_partial.data_ptr<float>(), nblocks,
+                         nrows, H, cur_stream());
+  rmsnorm_dw_reduce_launch(dw_partial.data_ptr<float>(), dw.data_ptr(),
+                           nblocks, H, cur_stream());
+  return {dx, dw};
 }
 
 // ---------------- fused qkv split + rope ----------------
@@ -286,6 +336,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("rope", &rope);
+  m.def("add_rmsnorm_fwd", &add_rmsnorm_fwd);
+  m.def("add_rmsnorm_bwd", &add_rmsnorm_bwd);
   m.def("qkv_rope_fwd", &qkv_rope_fwd);
   m.def("qkv_rope_bwd", &qkv_rope_bwd);
   m.def("silu_mul_fwd", &silu_mul_fwd);

@@ -190,4 +190,10 @@ void rmsnorm_bwd_launch(const void* dy, const void* x, const void* w,
   hipLaunchKernelGGL(rmsnorm_dw_reduce_kernel, dim3(rgrid), dim3(256), 0, s,
                      dw_partial, (short*)dw, nblocks, H);
 }
+void rmsnorm_dw_reduce_launch(const float* dw_partial, void* dw, int nblocks,
+                              int H, hipStream_t s) {
+  int rgrid = (H + 255) / 256;
+  hipLaunchKernelGGL(rmsnorm_dw_reduce_kernel, dim3(rgrid), dim3(256), 0, s,
+                     dw_partial, (short*)dw, nblocks, H);
+}
 }

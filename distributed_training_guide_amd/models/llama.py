@@ -23,6 +23,7 @@ import torch.nn.functional as F
 
 from ..ops import (RMSNorm, causal_lm_loss, flash_attention, qkv_rope,
                    silu_mul)
+from ..ops.rmsnorm import add_rmsnorm
 
 
 def pad_vocab(v: int, multiple: int = 64) -> int:
@@ -121,10 +122,25 @@ class LlamaDecoderLayer(nn.Module):
             config.hidden_size, config.rms_norm_eps, device, dtype)
         self.mlp = LlamaMLP(config, device, dtype)
 
-    def forward(self, x, position_ids=None):
-        x = x + self.self_attn(self.input_layernorm(x), position_ids)
-        x = x + self.mlp(self.post_attention_layernorm(x))
-        return x
+    def forward(self, x, position_ids=None, residual=None):
+        """Two call forms (both via __call__ so FSDP/checkpoint hooks fire):
+        - forward(x): the reference chain x + attn(norm(x)) + mlp(...).
+        - forward(delta, residual=res): fused (delta, residual) threading —
+          each residual add is fused into the next RMSNorm kernel; returns
+          (new_delta, new_residual), equivalent to forward(res + delta).
+        """
+        if residual is None:
+            x = x + self.self_attn(self.input_layernorm(x), position_ids)
+            x = x + self.mlp(self.post_attention_layernorm(x))
+            return x
+        normed, residual = add_rmsnorm(residual, x,
+                                       self.input_layernorm.weight,
+                                       self.input_layernorm.eps)
+        a = self.self_attn(normed, position_ids)
+        normed, residual = add_rmsnorm(residual, a,
+                                       self.post_attention_layernorm.weight,
+                                       self.post_attention_layernorm.eps)
+        return self.mlp(normed), residual
 
 
 class LlamaForCausalLM(nn.Module):
@@ -181,9 +197,22 @@ class LlamaForCausalLM(nn.Module):
         if position_ids is not None and position_ids.dim() == 2:
             position_ids = position_ids[0]
         x = self.embed_tokens(input_ids)
-        for layer in self.layers:
-            x = layer(x, position_ids)
-        x = self.norm(x)
+        if x.is_cuda:
+            # fused (delta, residual) path: every residual add is fused
+            # into the next RMSNorm kernel (add_rmsnorm.hip). Every layer
+            # is entered through __call__ so FSDP / activation-checkpoint
+            # hooks fire; the zero first delta costs one extra H-read.
+            residual = x
+            delta = torch.zeros_like(x)
+            for layer in self.layers:
+                delta, residual = layer(delta, position_ids,
+                                        residual=residual)
+            x, _ = add_rmsnorm(residual, delta, self.norm.weight,
+                               self.norm.eps)
+        else:
+            for layer in self.layers:
+                x = layer(x, position_ids)
+            x = self.norm(x)
         logits = self.lm_head(x)
         loss = None
         if labels is not None:

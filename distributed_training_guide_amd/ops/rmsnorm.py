@@ -49,3 +49,38 @@ class RMSNorm(torch.nn.Module):
 
     def extra_repr(self):
         return f"{self.weight.shape[0]}, eps={self.eps}"
+
+
+class _AddRMSNormFn(torch.autograd.Function):
+    """Fused residual-add + RMSNorm (add_rmsnorm.hip): res_out = res+delta,
+    y = rmsnorm(res_out)*w in one pass; backward folds the downstream
+    residual gradient into dx (no separate elementwise adds)."""
+
+    @staticmethod
+    def forward(ctx, res, delta, w, eps):
+        y, res_out, rstd = ext().add_rmsnorm_fwd(res, delta, w, eps)
+        ctx.save_for_backward(res_out, w, rstd)
+        return y, res_out
+
+    @staticmethod
+    def backward(ctx, dy, dres_out):
+        res_out, w, rstd = ctx.saved_tensors
+        if dres_out is None:
+            dres_out = torch.zeros_like(res_out)
+        dx, dw = ext().add_rmsnorm_bwd(dy.contiguous(),
+                                       dres_out.contiguous(), res_out, w,
+                                       rstd)
+        # d_res == d_delta == dx (the add distributes the gradient)
+        return dx, dx, dw, None
+
+
+def add_rmsnorm(res: torch.Tensor, delta: torch.Tensor, w: torch.Tensor,
+                eps: float = 1e-5):
+    """Returns (rmsnorm(res+delta)*w, res+delta)."""
+    H = res.shape[-1]
+    if use_hip(res, delta) and H % 8 == 0 and H <= 16384:
+        return _AddRMSNormFn.apply(res.contiguous(), delta.contiguous(), w,
+                                   eps)
+    s = res + delta
+    return rmsnorm_ref(s, w, eps) if not use_hip(res, delta) \
+        else rmsnorm(s, w, eps), s

@@ -47,3 +47,23 @@ def test_grad_norms_finite():
         g = p.grad.float()
         assert torch.isfinite(g).all(), f"non-finite grad in {n}"
         assert g.abs().max() < 1e3, f"exploding grad in {n}"
+
+
+def test_fused_residual_path_matches_cpu():
+    """GPU fused (delta, residual) decoder path vs CPU float32 reference
+    path: same weights, same batch, loss must agree to bf16 tolerance."""
+    from distributed_training_guide_amd.models import build_model
+
+    torch.manual_seed(7)
+    model = build_model("llama-debug", device=torch.device("cuda"),
+                        dtype=torch.bfloat16)
+    cpu = build_model("llama-debug", device=torch.device("cpu"),
+                      dtype=torch.float32)
+    cpu.load_state_dict({k: v.float().cpu()
+                         for k, v in model.state_dict().items()})
+    ids = torch.randint(0, 1024, (2, 96), device="cuda")
+    out = model(input_ids=ids, labels=ids)
+    out_cpu = cpu(input_ids=ids.cpu(), labels=ids.cpu())
+    rel = abs(out.loss.item() - out_cpu.loss.item()) / out_cpu.loss.item()
+    assert rel < 3e-2, f"fused GPU loss {out.loss.item()} vs CPU " \
+                       f"{out_cpu.loss.item()}"
