@@ -98,13 +98,6 @@ __device__ __forceinline__ uint32_t cvt_pk_bf16(float lo, float hi) {
   return r;
 }
 
-// transposed [D<=128][64] bf16 LDS image: index swizzled so the scalar
-// transposed staging writes spread banks while the per-row 8-col block
-// read stays 16B-contiguous.
-__device__ __forceinline__ int tr_idx(int d, int col) {
-  return d * 64 + (col ^ ((d & 7) << 3) ^ (((d >> 3) & 7) << 3));
-}
-
 // MFMA C-fragment "already-in-B/A-operand-order" row permutation: feeding
 // A-operand rows of tile mt in order perm16(mt, l15) makes C position
 // (mt, lg, r) hold source row (mt>>1)*32 + lg*8 + (mt&1)*4 + r, i.e. the
