@@ -145,27 +145,45 @@ __global__ void __launch_bounds__(256) rmsnorm_bwd_kernel(
   }
 }
 
-// reduce dw partials: [nblocks, H] f32 -> dw [H] bf16
-__global__ void __launch_bounds__(256) rmsnorm_dw_reduce_kernel(
-    const float* __restrict__ dw_partial, short* __restrict__ dw,
+// reduce dw partials: [nblocks, H] f32 -> dw [H] bf16.  Two stages so the
+// read of the (up to 2048 x H) partial buffer parallelizes over enough
+// blocks to reach memory bandwidth (a single H/256-block pass is
+// CU-bound); stage 1 folds rows into a [16][H] buffer, stage 2 finishes.
+#define DW_RY 16
+__global__ void __launch_bounds__(256) rmsnorm_dw_stage1_kernel(
+    const float* __restrict__ dw_partial, float* __restrict__ partial2,
     int nblocks, int H) {
+  const int y = blockIdx.y;
+  const int r0 = (int)(((int64_t)nblocks * y) / DW_RY);
+  const int r1 = (int)(((int64_t)nblocks * (y + 1)) / DW_RY);
   for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < H;
        i += gridDim.x * blockDim.x) {
-    // 8 independent accumulator chains: the row loop is latency-bound
-    // otherwise (one dependent L2/HBM load per iteration)
     float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    int b = 0;
-    for (; b + 8 <= nblocks; b += 8)
+    int b = r0;
+    for (; b + 8 <= r1; b += 8)
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         s[j] += dw_partial[(int64_t)(b + j) * H + i];
-    for (; b < nblocks; ++b) s[0] += dw_partial[(int64_t)b * H + i];
-    dw[i] = f2bf(((s[0] + s[1]) + (s[2] + s[3])) +
-                 ((s[4] + s[5]) + (s[6] + s[7])));
+    for (; b < r1; ++b) s[0] += dw_partial[(int64_t)b * H + i];
+    partial2[(int64_t)y * H + i] = ((s[0] + s[1]) + (s[2] + s[3])) +
+                                   ((s[4] + s[5]) + (s[6] + s[7]));
+  }
+}
+
+__global__ void __launch_bounds__(256) rmsnorm_dw_stage2_kernel(
+    const float* __restrict__ partial2, short* __restrict__ dw, int H) {
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < H;
+       i += gridDim.x * blockDim.x) {
+    float s = 0.f;
+#pragma unroll
+    for (int y = 0; y < DW_RY; ++y) s += partial2[(int64_t)y * H + i];
+    dw[i] = f2bf(s);
   }
 }
 
 extern "C" {
+void rmsnorm_dw_reduce_launch(const float*, void*, int, int,
+                              hipStream_t);
 void rmsnorm_fwd_launch(const void* x, const void* w, void* y, void* rstd,
                         int64_t nrows, int H, float eps, hipStream_t s) {
   int grid = (int)(nrows < 2048 ? nrows : 2048);
@@ -186,14 +204,20 @@ void rmsnorm_bwd_launch(const void* dy, const void* x, const void* w,
   hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(nblocks), dim3(256), shmem, s,
                      (const short*)dy, (const short*)x, (const short*)w,
                      (const float*)rstd, (short*)dx, dw_partial, nrows, H);
-  int rgrid = (H + 255) / 256;
-  hipLaunchKernelGGL(rmsnorm_dw_reduce_kernel, dim3(rgrid), dim3(256), 0, s,
-                     dw_partial, (short*)dw, nblocks, H);
+  rmsnorm_dw_reduce_launch(dw_partial, dw, nblocks, H, s);
 }
 void rmsnorm_dw_reduce_launch(const float* dw_partial, void* dw, int nblocks,
                               int H, hipStream_t s) {
+  // scratch for stage 1 lives at the END of dw_partial's allocation?  No —
+  // callers allocate exactly [nblocks, H]; reuse its first DW_RY rows is
+  // unsafe (still being read).  Grab a cached block via hipMallocAsync.
+  float* p2 = nullptr;
+  (void)hipMallocAsync((void**)&p2, (size_t)DW_RY * H * sizeof(float), s);
   int rgrid = (H + 255) / 256;
-  hipLaunchKernelGGL(rmsnorm_dw_reduce_kernel, dim3(rgrid), dim3(256), 0, s,
-                     dw_partial, (short*)dw, nblocks, H);
+  hipLaunchKernelGGL(rmsnorm_dw_stage1_kernel, dim3(rgrid, DW_RY), dim3(256),
+                     0, s, dw_partial, p2, nblocks, H);
+  hipLaunchKernelGGL(rmsnorm_dw_stage2_kernel, dim3(rgrid), dim3(256), 0, s,
+                     p2, (short*)dw, H);
+  (void)hipFreeAsync(p2, s);
 }
 }
