@@ -34,7 +34,8 @@ def parse_args():
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", default="llama-3-8b")
-    p.add_argument("--batch-size", type=int, default=16, help="per GPU")
+    p.add_argument("--batch-size", type=int, default=24, help="per GPU "
+                   "(204 GB HBM at bs24 + DDP buckets still fits 288 GB at N=8)")
     p.add_argument("--seq-length", type=int, default=1024)
     p.add_argument("--bucket-cap-mb", type=int, default=128)
     p.add_argument("--zero1", action="store_true",
