@@ -147,33 +147,42 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
   // staging: 16 lanes per q row, vectorized b128 writes (rm_idx swizzle:
   // conflict-free), one (h, q-tile) AHEAD so global latency hides behind
   // a full tile of MFMA
+  // LDS-DMA staging (global_load_lds; see attention_fwd.hip): the next
+  // (h, q-tile)'s Q/dO tiles stream into the other buffer with no staging
+  // registers and no vmcnt park; the per-lane global slot is pre-XOR'd so
+  // the contiguous landing equals the rm_idx-swizzled image.
   const int nslot = D >> 3;
-  const int srow0 = tid / 16;  // thread stages rows srow0 + {0,16,32,48}
-  const int sslot = tid & 15;
-  const bool svalid = sslot < nslot;
-  // T14 split staging: issue the next (h, q-tile)'s global loads into
-  // registers early (latency hides under this tile's MFMA stream), write
-  // them to LDS just before the barrier.
-  auto issue2 = [&](int it, const short* base0, bf16x8* out) {
+  const int rpc = 512 / D;
+  const int ncw = nslot >> 2;
+  const int srow_l = lane / nslot;
+  const int ssp = lane & (nslot - 1);
+  auto stage_qdo = [&](int it, int buf) {
     const int hh = hkv * group + it / nqt;
     const int qt = qstart + (it % nqt) * 64;
-    const short* bse = base0 + ((int64_t)b * S * Hq + hh) * D;
-    if (svalid) {
+    const short* qbse = q + ((int64_t)b * S * Hq + hh) * D;
+    const short* dbse = dout + ((int64_t)b * S * Hq + hh) * D;
 #pragma unroll
-      for (int vv = 0; vv < 4; ++vv) {
-        int r = qt + srow0 + vv * 16;
-        if (r >= S) r = S - 1;
-        out[vv] = *reinterpret_cast<const bf16x8*>(
-            bse + (int64_t)r * strideS_q + sslot * 8);
-      }
+    for (int i = 0; i < 4; ++i) {
+      if (i >= ncw) break;
+      const int ch = wid * ncw + i;
+      const int row = ch * rpc + srow_l;
+      int r = qt + row;
+      if (r >= S) r = S - 1;
+      const int gslot = (ssp ^ row) & (nslot - 1);
+      const int64_t goff = (int64_t)r * strideS_q + gslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(qbse +
+                                                                  goff),
+          (__attribute__((address_space(3))) unsigned int*)(q_lds[buf] +
+                                                            ch * 512),
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(dbse +
+                                                                  goff),
+          (__attribute__((address_space(3))) unsigned int*)(do_lds[buf] +
+                                                            ch * 512),
+          16, 0, 0);
     }
-  };
-  auto write2 = [&](short* dst, const bf16x8* vecs) {
-    if (!svalid) return;
-#pragma unroll
-    for (int vv = 0; vv < 4; ++vv)
-      *reinterpret_cast<bf16x8*>(
-          dst + rm_idx(srow0 + vv * 16, sslot * 8, D)) = vecs[vv];
   };
   auto stage_lse = [&](int it, int buf) {
     const int hh = hkv * group + it / nqt;
@@ -189,22 +198,14 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
     }
   };
 
-  {
-    bf16x8 qv[4], dv[4];
-    issue2(0, q, qv);
-    issue2(0, dout, dv);
-    write2(q_lds[0], qv);
-    write2(do_lds[0], dv);
-    stage_lse(0, 0);
-  }
+  stage_qdo(0, 0);
+  stage_lse(0, 0);
   __syncthreads();
 
   for (int it = 0; it < niter; ++it) {
     const int buf = it & 1;
-    const bool more = it + 1 < niter;
-    bf16x8 qnext[4];
-    if (more) {
-      issue2(it + 1, q, qnext);
+    if (it + 1 < niter) {
+      stage_qdo(it + 1, buf ^ 1);
       stage_lse(it + 1, buf ^ 1);
     }
 
@@ -251,12 +252,6 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
       pk_ds[kc][rp + 1] = cvt_pk_bf16(ds[2], ds[3]);
     }
 
-    // q write-back here, dO issue now: the two staging register windows
-    // never overlap (VGPR budget), and dO's latency hides under dV/dK
-    if (more) write2(q_lds[buf ^ 1], qnext);
-    bf16x8 donext[4];
-    if (more) issue2(it + 1, dout, donext);
-
     // ---- dV += P^T dO ; dK += dS^T Q (B operands via tr16 transpose
     // reads from the same row-major tiles) ----
 #pragma unroll
@@ -272,7 +267,6 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
                             dkacc[dt]);
       }
     }
-    if (more) write2(do_lds[buf ^ 1], donext);
     __syncthreads();
   }
 
@@ -378,21 +372,32 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
   const int kv_end = min(S, qtile * 128 + 128);
   const int ntiles = (kv_end + 63) / 64;
 
+  // LDS-DMA staging (global_load_lds; see attention_fwd.hip)
   const int nslot = D >> 3;
+  const int rpc = 512 / D;
+  const int ncw = nslot >> 2;
+  const int skey_l = lane / nslot;
+  const int ssp = lane & (nslot - 1);
   auto stage_k = [&](int kv0s, int buf) {
 #pragma unroll
-    for (int vv = 0; vv < 4; ++vv) {
-      int vecid = vv * 256 + tid;
-      int key = vecid / 16, slot = vecid & 15;
-      if (slot >= nslot || key >= 64) continue;
+    for (int i = 0; i < 4; ++i) {
+      if (i >= ncw) break;
+      const int ch = wid * ncw + i;
+      const int key = ch * rpc + skey_l;
       int kg = kv0s + key;
       if (kg >= S) kg = S - 1;
-      bf16x8 kv_ = *reinterpret_cast<const bf16x8*>(
-          kb + (int64_t)kg * strideS_kv + slot * 8);
-      bf16x8 vv_ = *reinterpret_cast<const bf16x8*>(
-          vb + (int64_t)kg * strideS_kv + slot * 8);
-      *reinterpret_cast<bf16x8*>(k_lds[buf] + rm_idx(key, slot * 8, D)) = kv_;
-      *reinterpret_cast<bf16x8*>(v_lds[buf] + rm_idx(key, slot * 8, D)) = vv_;
+      const int gslot = (ssp ^ key) & (nslot - 1);
+      const int64_t goff = (int64_t)kg * strideS_kv + gslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(kb + goff),
+          (__attribute__((address_space(3))) unsigned int*)(k_lds[buf] +
+                                                            ch * 512),
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(vb + goff),
+          (__attribute__((address_space(3))) unsigned int*)(v_lds[buf] +
+                                                            ch * 512),
+          16, 0, 0);
     }
   };
 

@@ -147,25 +147,37 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
   const float c = scale * LOG2E;
   (void)hkv;
 
-  // ---- staging: K and V row-major [key][d], rm_idx-swizzled, 16 lanes
-  // per key row (conflict-free vectorized b128 writes, coalesced global
-  // reads); loaded ONE TILE AHEAD so the global latency lands behind a
-  // full tile of MFMA (T14) ----
+  // ---- staging via LDS-DMA (global_load_lds, probed: data lands at
+  // lds_base + lane*16 with per-lane global sources — tools/gldsprobe.hip):
+  // no staging registers, no ds_write pass, no vmcnt park.  The per-lane
+  // global slot is pre-XOR'd so the contiguous landing equals the
+  // rm_idx-swizzled image; tile t+1's DMA issues at the top of tile t and
+  // completes under a full tile of MFMA before the end-of-tile barrier.
   const int nslot = D >> 3;
+  const int rpc = 512 / D;        // rows per 1 KiB chunk (64 lanes x 16 B)
+  const int ncw = nslot >> 2;     // chunks per wave per tensor
+  const int skey_l = lane / nslot;
+  const int ssp = lane & (nslot - 1);
   auto stage_kv = [&](int kv0, int buf) {
 #pragma unroll
-    for (int vv = 0; vv < 4; ++vv) {
-      int vecid = vv * 256 + tid;
-      int key = vecid / 16, slot = vecid & 15;
-      if (slot >= nslot || key >= KVBLK) continue;
+    for (int i = 0; i < 4; ++i) {
+      if (i >= ncw) break;
+      const int ch = wid * ncw + i;
+      const int key = ch * rpc + skey_l;
       int kg = kv0 + key;
       if (kg >= S) kg = S - 1;
-      bf16x8 kv_ = *reinterpret_cast<const bf16x8*>(
-          kb + (int64_t)kg * strideS_kv + slot * 8);
-      bf16x8 vv_ = *reinterpret_cast<const bf16x8*>(
-          vb + (int64_t)kg * strideS_kv + slot * 8);
-      *reinterpret_cast<bf16x8*>(k_lds[buf] + rm_idx(key, slot * 8, D)) = kv_;
-      *reinterpret_cast<bf16x8*>(v_lds[buf] + rm_idx(key, slot * 8, D)) = vv_;
+      const int gslot = (ssp ^ key) & (nslot - 1);
+      const int64_t goff = (int64_t)kg * strideS_kv + gslot * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(kb + goff),
+          (__attribute__((address_space(3))) unsigned int*)(k_lds[buf] +
+                                                            ch * 512),
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(vb + goff),
+          (__attribute__((address_space(3))) unsigned int*)(v_lds[buf] +
+                                                            ch * 512),
+          16, 0, 0);
     }
   };
 
