@@ -34,9 +34,16 @@ def main():
           f"resuming at iteration {state['iteration']}")
 
     fail_prob = float(os.environ.get("TOY_FAIL_PROB", "0.05"))
+    # deterministic injection for tests: rank 0 fails once at this iter on
+    # the first incarnation (TOY_FAIL_AT=-1 disables)
+    fail_at = int(os.environ.get("TOY_FAIL_AT", "-1"))
+    restarts = int(os.environ.get("TORCHELASTIC_RESTART_COUNT", 0))
     for it in range(state["iteration"], 20):
         time.sleep(0.1)  # "training"
-        if random.random() < fail_prob:
+        if fail_at >= 0:
+            if rank == 0 and restarts == 0 and it == fail_at:
+                raise RuntimeError(f"rank {rank} injected failure at {it}")
+        elif random.random() < fail_prob:
             raise RuntimeError(f"rank {rank} simulated failure at iter {it}")
         state["iteration"] = it + 1
         dist.barrier()

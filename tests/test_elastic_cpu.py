@@ -16,11 +16,11 @@ REPO = Path(__file__).resolve().parent.parent
 def test_toy_elastic_restart(tmp_path):
     env = dict(os.environ)
     env["TOY_STATE_FILE"] = str(tmp_path / "toy-state.json")
-    env["TOY_FAIL_PROB"] = "0.05"
+    env["TOY_FAIL_AT"] = "3"  # deterministic: rank 0 dies once at iter 3
     proc = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--standalone",
          "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
-         "--max-restarts", "20",
+         "--max-restarts", "3",
          str(REPO / "related-topics" / "elastic-training" / "toy.py")],
         env=env, cwd=str(tmp_path), capture_output=True, text=True,
         timeout=280)
@@ -37,6 +37,7 @@ def test_toy_no_failures(tmp_path):
     env = dict(os.environ)
     env["TOY_STATE_FILE"] = str(tmp_path / "toy-state.json")
     env["TOY_FAIL_PROB"] = "0"
+    env["TOY_FAIL_AT"] = "-1"
     proc = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--standalone",
          "--local-addr", "127.0.0.1", "--nproc-per-node", "2",
