@@ -220,6 +220,7 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
     const int key = kv0 + l15;  // this lane's key (C n-position)
     const bool diag = (qt < kt * 64 + 64) || (qt + 63 >= S);
     u32x4 pk_p[2], pk_ds[2];
+    __builtin_amdgcn_s_setprio(1);  // T5: prioritize the MFMA stream
 #pragma unroll
     for (int mt = 0; mt < 4; ++mt) {
       const int qr = perm16(mt, l15);  // local row in the staged tile
@@ -267,6 +268,7 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
                             dkacc[dt]);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
 
@@ -424,6 +426,7 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
       u32x4 pk_ds[2][2];  // [nq][kc]
       const short* kl = k_lds[buf];
       const short* vl = v_lds[buf];
+      __builtin_amdgcn_s_setprio(1);  // T5: prioritize the MFMA stream
 #pragma unroll
       for (int mt = 0; mt < 4; ++mt) {
         const int kr = perm16(mt, l15);  // local key row
@@ -485,6 +488,7 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
                                  kfr, dqacc[dt][1]);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
     __syncthreads();
   }

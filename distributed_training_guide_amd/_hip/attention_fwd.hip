@@ -202,6 +202,7 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
     // row-major staged tile (ds_read_b128, ~50cyc, hidden by MFMA) ----
     const short* kl = k_lds[buf];
     f32x4 sfrag[4][2];
+    __builtin_amdgcn_s_setprio(1);  // T5: prioritize the QK^T MFMA stream
 #pragma unroll
     for (int mt = 0; mt < 4; ++mt) {
       const int kr = kperm[mt];  // local key row (permuted)
@@ -217,6 +218,7 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
       sfrag[mt][0] = acc0;
       sfrag[mt][1] = acc1;
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // ---- causal mask + online softmax (rows lane-local) ----
     u32x4 pk[2][2];  // [nq][kc]: 4 regs of 2 bf16 (keys j=0..7)
@@ -271,6 +273,7 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
     // ---- O^T += mfma(V^T, P^T): V^T A-frags via tr16 hardware-transpose
     // reads from the row-major V tile (no transposed copy) ----
     const short* vt = v_lds[buf];
+    __builtin_amdgcn_s_setprio(1);  // T5: prioritize the P.V MFMA stream
 #pragma unroll
     for (int dt = 0; dt < 8; ++dt) {
       if (dt >= nd16) break;
@@ -288,6 +291,7 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
         oacc[dt][nq] = acc;
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
 
