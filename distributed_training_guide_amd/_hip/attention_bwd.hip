@@ -73,7 +73,7 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
     short* __restrict__ dk, short* __restrict__ dv, int B, int S, int Hq,
     int Hkv, int D, float scale) {
   // double-buffered ROW-major staging of the 64-q-row tile (q[qrow][d],
-  // dO[qrow][d], rm_idx-swizzled): serves BOTH the Q/dO A-fragments
+  // dO[qrow][d], st_idx subtile image): serves BOTH the Q/dO A-fragments
   // (b128 row reads, replacing per-tile global gathers) and the
   // dV/dK B-fragments (tr16 hardware-transpose reads); plus lse/delta rows
   __shared__ short q_lds[2][64 * 128];
@@ -144,18 +144,17 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
   const int nqt = (S - qstart + 63) / 64;
   const int niter = group * nqt;
 
-  // staging: 16 lanes per q row, vectorized b128 writes (rm_idx swizzle:
+  // staging via LDS-DMA (st_idx subtile image;
   // conflict-free), one (h, q-tile) AHEAD so global latency hides behind
   // a full tile of MFMA
   // LDS-DMA staging (global_load_lds; see attention_fwd.hip): the next
   // (h, q-tile)'s Q/dO tiles stream into the other buffer with no staging
   // registers and no vmcnt park; the per-lane global slot is pre-XOR'd so
-  // the contiguous landing equals the rm_idx-swizzled image.
+  // the contiguous landing equals the st_idx subtile image.
   const int nslot = D >> 3;
-  const int rpc = 512 / D;
   const int ncw = nslot >> 2;
-  const int srow_l = lane / nslot;
-  const int ssp = lane & (nslot - 1);
+  const int srow_half = lane >> 1;       // subtile-half row within chunk
+  const int scol_half = (lane & 1) * 8;  // 8-col half within subtile
   auto stage_qdo = [&](int it, int buf) {
     const int hh = hkv * group + it / nqt;
     const int qt = qstart + (it % nqt) * 64;
@@ -165,11 +164,11 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
     for (int i = 0; i < 4; ++i) {
       if (i >= ncw) break;
       const int ch = wid * ncw + i;
-      const int row = ch * rpc + srow_l;
+      const int row = (ch & 1) * 32 + srow_half;
       int r = qt + row;
       if (r >= S) r = S - 1;
-      const int gslot = (ssp ^ row) & (nslot - 1);
-      const int64_t goff = (int64_t)r * strideS_q + gslot * 8;
+      const int64_t goff =
+          (int64_t)r * strideS_q + (ch >> 1) * 16 + scol_half;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(qbse +
                                                                   goff),
@@ -229,9 +228,9 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
       for (int c = 0; c < 4; ++c)
         if (c < nkc) {
           bf16x8 qa = *reinterpret_cast<const bf16x8*>(
-              ql + rm_idx(qr, c * 32 + lg * 8, D));
+              ql + st_idx(qr, c * 32 + lg * 8));
           bf16x8 doa = *reinterpret_cast<const bf16x8*>(
-              dol + rm_idx(qr, c * 32 + lg * 8, D));
+              dol + st_idx(qr, c * 32 + lg * 8));
           sa = mfma16b(qa, kf[c], sa);
           da = mfma16b(doa, vf[c], da);
         }
@@ -260,8 +259,8 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
       if (dt >= nd16) break;
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
-        bf16x8 dofr = tr16_frag(dol, kc * 32 + lg * 8, dt * 16, D, l15);
-        bf16x8 qfr = tr16_frag(ql, kc * 32 + lg * 8, dt * 16, D, l15);
+        bf16x8 dofr = tr16_frag_st(dol, kc * 32 + lg * 8, dt * 16, l15);
+        bf16x8 qfr = tr16_frag_st(ql, kc * 32 + lg * 8, dt * 16, l15);
         dvacc[dt] = mfma16b(__builtin_bit_cast(bf16x8, pk_p[kc]), dofr,
                             dvacc[dt]);
         dkacc[dt] = mfma16b(__builtin_bit_cast(bf16x8, pk_ds[kc]), qfr,
@@ -296,7 +295,7 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dq, int B, int S, int Hq, int Hkv, int D,
     float scale) {
-  // double-buffered ROW-major K and V tiles (rm_idx-swizzled): K serves
+  // double-buffered K and V tiles (st_idx subtile image): K serves
   // the S^T A-fragments (b128 row reads) AND the dQ B-fragments (tr16
   // transpose reads); V serves the dP^T A-fragments
   __shared__ short k_lds[2][64 * 128];
@@ -376,20 +375,19 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
 
   // LDS-DMA staging (global_load_lds; see attention_fwd.hip)
   const int nslot = D >> 3;
-  const int rpc = 512 / D;
   const int ncw = nslot >> 2;
-  const int skey_l = lane / nslot;
-  const int ssp = lane & (nslot - 1);
+  const int srow_half = lane >> 1;       // subtile-half row within chunk
+  const int scol_half = (lane & 1) * 8;  // 8-col half within subtile
   auto stage_k = [&](int kv0s, int buf) {
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       if (i >= ncw) break;
       const int ch = wid * ncw + i;
-      const int key = ch * rpc + skey_l;
+      const int key = (ch & 1) * 32 + srow_half;
       int kg = kv0s + key;
       if (kg >= S) kg = S - 1;
-      const int gslot = (ssp ^ key) & (nslot - 1);
-      const int64_t goff = (int64_t)kg * strideS_kv + gslot * 8;
+      const int64_t goff =
+          (int64_t)kg * strideS_kv + (ch >> 1) * 16 + scol_half;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(kb + goff),
           (__attribute__((address_space(3))) unsigned int*)(k_lds[buf] +
@@ -436,9 +434,9 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
         for (int c = 0; c < 4; ++c)
           if (c < nkc) {
             bf16x8 ka = *reinterpret_cast<const bf16x8*>(
-                kl + rm_idx(kr, c * 32 + lg * 8, D));
+                kl + st_idx(kr, c * 32 + lg * 8));
             bf16x8 va = *reinterpret_cast<const bf16x8*>(
-                vl + rm_idx(kr, c * 32 + lg * 8, D));
+                vl + st_idx(kr, c * 32 + lg * 8));
             s0 = mfma16b(ka, qf[0][c], s0);
             d0 = mfma16b(va, dof[0][c], d0);
             s1 = mfma16b(ka, qf[1][c], s1);
@@ -478,9 +476,9 @@ __global__ void __launch_bounds__(256, 2) attn_dq_kernel(
           bf16x8 kfr;
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            kfr[j] = kl[rm_idx(kc * 32 + lg * 8 + j, dt * 16 + l15, D)];
+            kfr[j] = kl[st_idx(kc * 32 + lg * 8 + j, dt * 16 + l15)];
 #else
-          bf16x8 kfr = tr16_frag(kl, kc * 32 + lg * 8, dt * 16, D, l15);
+          bf16x8 kfr = tr16_frag_st(kl, kc * 32 + lg * 8, dt * 16, l15);
 #endif
           dqacc[dt][0] = mfma16b(__builtin_bit_cast(bf16x8, pk_ds[0][kc]),
                                  kfr, dqacc[dt][0]);

@@ -56,7 +56,8 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
     const short* __restrict__ v, short* __restrict__ o,
     float* __restrict__ lse_out, int B, int S, int Hq, int Hkv, int D,
     float scale) {
-  // both tiles row-major [key][d], rm_idx slot-swizzled, double-buffered
+  // both tiles in the 16-col-subtile image (st_idx; conflict-free tr16
+  // reads + b128 row reads), double-buffered
   __shared__ short k_lds[2][KVBLK * 128];
   __shared__ short v_lds[2][KVBLK * 128];
 
@@ -151,23 +152,22 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
   // lds_base + lane*16 with per-lane global sources — tools/gldsprobe.hip):
   // no staging registers, no ds_write pass, no vmcnt park.  The per-lane
   // global slot is pre-XOR'd so the contiguous landing equals the
-  // rm_idx-swizzled image; tile t+1's DMA issues at the top of tile t and
+  // st_idx subtile image; tile t+1's DMA issues at the top of tile t and
   // completes under a full tile of MFMA before the end-of-tile barrier.
   const int nslot = D >> 3;
-  const int rpc = 512 / D;        // rows per 1 KiB chunk (64 lanes x 16 B)
   const int ncw = nslot >> 2;     // chunks per wave per tensor
-  const int skey_l = lane / nslot;
-  const int ssp = lane & (nslot - 1);
+  const int srow_half = lane >> 1;       // subtile-half row within chunk
+  const int scol_half = (lane & 1) * 8;  // 8-col half within subtile
   auto stage_kv = [&](int kv0, int buf) {
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       if (i >= ncw) break;
       const int ch = wid * ncw + i;
-      const int key = ch * rpc + skey_l;
+      const int key = (ch & 1) * 32 + srow_half;
       int kg = kv0 + key;
       if (kg >= S) kg = S - 1;
-      const int gslot = (ssp ^ key) & (nslot - 1);
-      const int64_t goff = (int64_t)kg * strideS_kv + gslot * 8;
+      const int64_t goff =
+          (int64_t)kg * strideS_kv + (ch >> 1) * 16 + scol_half;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(kb + goff),
           (__attribute__((address_space(3))) unsigned int*)(k_lds[buf] +
@@ -211,7 +211,7 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
       for (int kc = 0; kc < 4; ++kc)
         if (kc < nkc) {
           bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-              kl + rm_idx(kr, kc * 32 + lg * 8, D));
+              kl + st_idx(kr, kc * 32 + lg * 8));
           acc0 = mfma16(kf, qf[0][kc], acc0);
           acc1 = mfma16(kf, qf[1][kc], acc1);
         }
@@ -280,7 +280,7 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
       bf16x8 va[2];
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc)
-        va[kc] = tr16_frag(vt, kc * 32 + lg * 8, dt * 16, D, l15);
+        va[kc] = tr16_frag_st(vt, kc * 32 + lg * 8, dt * 16, l15);
 #pragma unroll
       for (int nq = 0; nq < 2; ++nq) {
         f32x4 acc = oacc[dt][nq];

@@ -148,3 +148,29 @@ __device__ __forceinline__ s16x8 tr16_frag(const short* lds_base, int row0,
   }
   return out;
 }
+
+// 16-col-subtile image ([64 rows][16 cols] blocks, row stride 32 B): the
+// guide's conflict-free layout for ds_read_b64_tr_b16, still b128-readable
+// per row (8-col runs at col%16 in {0,8}). Pairs with the glds source
+// mapping in the attention kernels (chunk ch lands rows (ch&1)*32 + l/2,
+// cols (ch>>1)*16 + (l&1)*8).
+__device__ __forceinline__ int st_idx(int row, int col) {
+  return ((col >> 4) << 10) + (row << 4) + (col & 15);
+}
+
+__device__ __forceinline__ s16x8 tr16_frag_st(const short* lds_base, int row0,
+                                              int col0, int l15) {
+  const int r = row0 + (l15 >> 2);
+  const int c = col0 + (l15 & 3) * 4;
+  s16x4tr lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (LDS_AS s16x4tr*)(lds_base + st_idx(r, c)));
+  s16x4tr hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (LDS_AS s16x4tr*)(lds_base + st_idx(r + 4, c)));
+  s16x8 out;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    out[j] = lo[j];
+    out[4 + j] = hi[j];
+  }
+  return out;
+}
