@@ -36,25 +36,29 @@ __device__ __forceinline__ f32x4 mfma16b(bf16x8 a, bf16x8 b, f32x4 c) {
 }
 
 // ---------------- delta = rowsum(dO * O) ----------------
+// one row per 16-lane group (16 lanes x 8 bf16 = 128 elems = a whole
+// D=128 row per pass) — a row-per-wave version left 48/64 lanes idle at
+// D<=128 (guide §5 mistake 6)
 __global__ void __launch_bounds__(256) attn_delta_kernel(
     const short* __restrict__ dout, const short* __restrict__ o,
     float* __restrict__ delta, int B, int S, int Hq, int D) {
-  const int wid = threadIdx.x / WAVE;
-  const int lane = threadIdx.x & (WAVE - 1);
+  const int grp = threadIdx.x >> 4;       // 16 groups per block
+  const int gl = threadIdx.x & 15;
   int64_t nrows = (int64_t)B * S * Hq;
-  for (int64_t row = (int64_t)blockIdx.x * 4 + wid; row < nrows;
-       row += (int64_t)gridDim.x * 4) {
+  for (int64_t row = (int64_t)blockIdx.x * 16 + grp; row < nrows;
+       row += (int64_t)gridDim.x * 16) {
     const short* dp = dout + row * D;
     const short* op = o + row * D;
     float s = 0.f;
-    for (int i = lane * 8; i < D; i += WAVE * 8) {
+    for (int i = gl * 8; i < D; i += 16 * 8) {
       bf16x8 dv = *reinterpret_cast<const bf16x8*>(dp + i);
       bf16x8 ov = *reinterpret_cast<const bf16x8*>(op + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j) s += bf2f(dv[j]) * bf2f(ov[j]);
     }
-    s = wave_reduce_sum(s);
-    if (lane == 0) {
+#pragma unroll
+    for (int off = 1; off < 16; off <<= 1) s += __shfl_xor(s, off, WAVE);
+    if (gl == 0) {
       // row = (b*S + s_pos)*Hq + h  ->  delta is [B, Hq, S]
       int h = (int)(row % Hq);
       int64_t bs = row / Hq;
@@ -514,7 +518,7 @@ void attn_bwd_launch(const void* dout, const void* q, const void* k,
                      float* delta, void* dq, void* dk, void* dv, int B, int S,
                      int Hq, int Hkv, int D, float scale, hipStream_t stream) {
   int64_t nrows = (int64_t)B * S * Hq;
-  int64_t dwant = (nrows + 3) / 4;
+  int64_t dwant = (nrows + 15) / 16;
   int dgrid = (int)(dwant < 2048 ? (dwant < 1 ? 1 : dwant) : 2048);
   hipLaunchKernelGGL(attn_delta_kernel, dim3(dgrid), dim3(256), 0, stream,
                      (const short*)dout, (const short*)o, delta, B, S, Hq, D);
