@@ -41,7 +41,7 @@ __global__ void __launch_bounds__(256) add_rmsnorm_fwd_kernel(
       *reinterpret_cast<s16x8*>(ro + i) = sv;
       *reinterpret_cast<s16x8*>(rowbuf + i) = sv;
     }
-    ss = block_reduce_sum(ss, scratch);
+    ss = block4_sum(ss, scratch);
     const float rstd = rsqrtf(ss * invH + eps);
     if (threadIdx.x == 0 && rstd_out != nullptr) rstd_out[row] = rstd;
     for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
@@ -91,7 +91,7 @@ __global__ void __launch_bounds__(256) add_rmsnorm_bwd_kernel(
       for (int j = 0; j < 8; ++j)
         dot += bf2f(dv[j]) * bf2f(wv[j]) * bf2f(xv[j]) * rs;
     }
-    dot = block_reduce_sum(dot, scratch) * invH;
+    dot = block4_sum(dot, scratch) * invH;
     int c = 0;
     for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8, ++c) {
       s16x8 dv = *reinterpret_cast<const s16x8*>(dy_l + i);

@@ -84,6 +84,19 @@ __device__ __forceinline__ float block_reduce_max(float v, float* scratch) {
   return out;
 }
 
+
+// 4-wave block sum with ONE barrier (blockDim == 256): wave partials to
+// scratch, one sync, every thread folds the 4 partials. The caller's
+// loop-end __syncthreads() protects scratch reuse across iterations.
+__device__ __forceinline__ float block4_sum(float v, float* scratch) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  v = wave_reduce_sum(v);
+  if (lane == 0) scratch[wid] = v;
+  __syncthreads();
+  return (scratch[0] + scratch[1]) + (scratch[2] + scratch[3]);
+}
+
 #define HIP_CHECK_DEV(expr)                                                  \
   do {                                                                       \
     hipError_t _e = (expr);                                                  \

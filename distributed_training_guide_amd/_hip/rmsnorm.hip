@@ -32,7 +32,7 @@ __global__ void __launch_bounds__(256) rmsnorm_fwd_kernel(
         float f = bf2f(xr[i]); ss += f * f;
       }
     }
-    ss = block_reduce_sum(ss, scratch);
+    ss = block4_sum(ss, scratch);
     const float rstd = rsqrtf(ss * invH + eps);
     if (threadIdx.x == 0 && rstd_out != nullptr) rstd_out[row] = rstd;
     if ((H & 7) == 0) {
@@ -100,7 +100,7 @@ __global__ void __launch_bounds__(256) rmsnorm_bwd_kernel(
         dot += bf2f(dyr[i]) * bf2f(w[i]) * bf2f(xr[i]) * rs;
       }
     }
-    dot = block_reduce_sum(dot, scratch) * invH;
+    dot = block4_sum(dot, scratch) * invH;
     if (vec) {
       int c = 0;
       for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8, ++c) {
