@@ -63,8 +63,10 @@ def _fsdp_train_and_compare(rank, world):
             f"{n} diff {(full[n] - pr.detach()).abs().max()}"
 
 
-def test_fsdp_matches_single_process():
-    run_dist(_fsdp_train_and_compare, world_size=2)
+@pytest.mark.parametrize("world", [2, 3])
+def test_fsdp_matches_single_process(world):
+    # world=3: shard padding (param counts not divisible by world)
+    run_dist(_fsdp_train_and_compare, world_size=world)
 
 
 def _fsdp_meta_init(rank, world):

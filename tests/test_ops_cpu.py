@@ -99,3 +99,19 @@ def test_cross_entropy_ref_ignore():
     labels[0, 3] = -100
     loss = R.cross_entropy_ref(logits, labels)
     assert torch.isfinite(loss)
+
+
+def test_add_rmsnorm_cpu_fallback():
+    """Fused residual-add + rmsnorm CPU path == manual add + rmsnorm."""
+    import torch
+
+    from distributed_training_guide_amd.ops.rmsnorm import add_rmsnorm
+    from distributed_training_guide_amd.ops.reference import rmsnorm_ref
+
+    torch.manual_seed(0)
+    res = torch.randn(2, 8, 64)
+    delta = torch.randn(2, 8, 64)
+    w = torch.randn(64)
+    y, res_out = add_rmsnorm(res, delta, w, 1e-5)
+    assert torch.allclose(res_out, res + delta)
+    assert torch.allclose(y, rmsnorm_ref(res + delta, w, 1e-5), atol=1e-5)
