@@ -61,6 +61,10 @@ def get_parser(extra: bool = True) -> argparse.ArgumentParser:
         p.add_argument("--device", default=None,
                        help="override device (cpu/cuda); default: auto")
         p.add_argument("--num-workers", default=1, type=int)
+        p.add_argument("--measure-waiting", action="store_true",
+                       help="time an explicit barrier before each data "
+                            "fetch to separate straggler waiting from "
+                            "loading (optimizing-data-loading recipe)")
     return p
 
 
@@ -139,8 +143,12 @@ def run_training(args, strategy):
 
     wandb_run = _maybe_init_wandb(args, strategy, resumed)
 
-    timers = {k: LocalTimer(strategy.device)
-              for k in ["data", "forward", "backward", "update"]}
+    phases = ["data", "forward", "backward", "update"]
+    measure_waiting = (getattr(args, "measure_waiting", False)
+                       and strategy.world_size > 1)
+    if measure_waiting:
+        phases = ["waiting"] + phases
+    timers = {k: LocalTimer(strategy.device) for k in phases}
     accum = max(1, getattr(args, "grad_accum_steps", 1))
     max_steps = getattr(args, "max_steps", 0)
     tok_per_step = args.batch_size * args.seq_length * strategy.dp_size * accum
@@ -157,6 +165,13 @@ def run_training(args, strategy):
         n_batches = len(dataloader)
 
         for i_step in range(n_batches // accum):
+            if measure_waiting:
+                # all ranks rendezvous BEFORE fetching: time spent here is
+                # waiting on stragglers, not on this rank's loader
+                import torch.distributed as dist
+
+                with timers["waiting"]:
+                    dist.barrier()
             micro = []
             with timers["data"], torch.no_grad():
                 for _ in range(accum):
