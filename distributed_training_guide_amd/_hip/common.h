@@ -134,33 +134,6 @@ typedef __attribute__((ext_vector_type(4))) short s16x4tr;
 typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 #define LDS_AS __attribute__((address_space(3)))
 
-// row-major [rows][DCOLS] bf16 image with 16B-slot XOR swizzle (slot ^=
-// row & (nslot-1)): conflict-free vectorized staging writes (16 lanes per
-// row) and b128 row reads; tr16 reads compute the swizzled slot per lane.
-__device__ __forceinline__ int rm_idx(int row, int col, int ncols) {
-  const int nslot1 = (ncols >> 3) - 1;
-  return row * ncols + ((((col >> 3) ^ row) & nslot1) << 3) + (col & 7);
-}
-
-// B/A-operand fragment (k-dim = 8 consecutive rows at this lane's column)
-// from a row-major swizzled image: two tr16 reads, rows row0..+7, column
-// col0 + l15.
-__device__ __forceinline__ s16x8 tr16_frag(const short* lds_base, int row0,
-                                           int col0, int ncols, int l15) {
-  const int r = row0 + (l15 >> 2);
-  const int c = col0 + (l15 & 3) * 4;
-  s16x4tr lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-      (LDS_AS s16x4tr*)(lds_base + rm_idx(r, c, ncols)));
-  s16x4tr hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-      (LDS_AS s16x4tr*)(lds_base + rm_idx(r + 4, c, ncols)));
-  s16x8 out;
-#pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    out[j] = lo[j];
-    out[4 + j] = hi[j];
-  }
-  return out;
-}
 
 // 16-col-subtile image ([64 rows][16 cols] blocks, row stride 32 B): the
 // guide's conflict-free layout for ds_read_b64_tr_b16, still b128-readable

@@ -38,7 +38,8 @@ def cpos16(mt, lg):
 
 
 def rm_idx(row, col):
-    return row * D + ((((col >> 3) ^ row) & nslot1) << 3) + (col & 7)
+    # st_idx: 16-col-subtile image (matches _hip/common.h st_idx)
+    return ((col >> 4) << 10) + (row << 4) + (col & 15)
 
 
 def mfma(A, B, C):
