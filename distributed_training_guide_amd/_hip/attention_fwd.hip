@@ -204,19 +204,29 @@ __global__ void __launch_bounds__(256, 2) attn_fwd_kernel(
     f32x4 sfrag[4][2];
     __builtin_amdgcn_s_setprio(1);  // T5: prioritize the QK^T MFMA stream
 #pragma unroll
-    for (int mt = 0; mt < 4; ++mt) {
-      const int kr = kperm[mt];  // local key row (permuted)
-      f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
+    for (int mth = 0; mth < 2; ++mth) {
+      // two key-tiles at once -> 4 independent accumulator chains (the
+      // dependent-accumulator MFMA latency exceeds the issue rate; a
+      // 2-chain version left the pipe half idle)
+      const int kra = kperm[2 * mth], krb = kperm[2 * mth + 1];
+      f32x4 a0 = {0.f, 0.f, 0.f, 0.f}, a1 = {0.f, 0.f, 0.f, 0.f};
+      f32x4 b0 = {0.f, 0.f, 0.f, 0.f}, b1 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int kc = 0; kc < 4; ++kc)
         if (kc < nkc) {
-          bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-              kl + st_idx(kr, kc * 32 + lg * 8));
-          acc0 = mfma16(kf, qf[0][kc], acc0);
-          acc1 = mfma16(kf, qf[1][kc], acc1);
+          bf16x8 kfa = *reinterpret_cast<const bf16x8*>(
+              kl + st_idx(kra, kc * 32 + lg * 8));
+          bf16x8 kfb = *reinterpret_cast<const bf16x8*>(
+              kl + st_idx(krb, kc * 32 + lg * 8));
+          a0 = mfma16(kfa, qf[0][kc], a0);
+          b0 = mfma16(kfb, qf[0][kc], b0);
+          a1 = mfma16(kfa, qf[1][kc], a1);
+          b1 = mfma16(kfb, qf[1][kc], b1);
         }
-      sfrag[mt][0] = acc0;
-      sfrag[mt][1] = acc1;
+      sfrag[2 * mth][0] = a0;
+      sfrag[2 * mth][1] = a1;
+      sfrag[2 * mth + 1][0] = b0;
+      sfrag[2 * mth + 1][1] = b1;
     }
     __builtin_amdgcn_s_setprio(0);
 
