@@ -225,35 +225,50 @@ __global__ void __launch_bounds__(256, 2) attn_dkdv_kernel(
     u32x4 pk_p[2], pk_ds[2];
     __builtin_amdgcn_s_setprio(1);  // T5: prioritize the MFMA stream
 #pragma unroll
-    for (int mt = 0; mt < 4; ++mt) {
-      const int qr = perm16(mt, l15);  // local row in the staged tile
-      f32x4 sa = {0.f, 0.f, 0.f, 0.f}, da = {0.f, 0.f, 0.f, 0.f};
+    for (int mth = 0; mth < 2; ++mth) {
+      // two q-row tiles at once -> 4 independent MFMA accumulator chains
+      // (dependent-accumulator latency > issue rate with only 2)
+      const int qra = perm16(2 * mth, l15), qrb = perm16(2 * mth + 1, l15);
+      f32x4 sa0 = {0.f, 0.f, 0.f, 0.f}, da0 = {0.f, 0.f, 0.f, 0.f};
+      f32x4 sa1 = {0.f, 0.f, 0.f, 0.f}, da1 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int c = 0; c < 4; ++c)
         if (c < nkc) {
-          bf16x8 qa = *reinterpret_cast<const bf16x8*>(
-              ql + st_idx(qr, c * 32 + lg * 8));
-          bf16x8 doa = *reinterpret_cast<const bf16x8*>(
-              dol + st_idx(qr, c * 32 + lg * 8));
-          sa = mfma16b(qa, kf[c], sa);
-          da = mfma16b(doa, vf[c], da);
+          bf16x8 qa0 = *reinterpret_cast<const bf16x8*>(
+              ql + st_idx(qra, c * 32 + lg * 8));
+          bf16x8 qa1 = *reinterpret_cast<const bf16x8*>(
+              ql + st_idx(qrb, c * 32 + lg * 8));
+          bf16x8 do0 = *reinterpret_cast<const bf16x8*>(
+              dol + st_idx(qra, c * 32 + lg * 8));
+          bf16x8 do1 = *reinterpret_cast<const bf16x8*>(
+              dol + st_idx(qrb, c * 32 + lg * 8));
+          sa0 = mfma16b(qa0, kf[c], sa0);
+          sa1 = mfma16b(qa1, kf[c], sa1);
+          da0 = mfma16b(do0, vf[c], da0);
+          da1 = mfma16b(do1, vf[c], da1);
         }
-      const int qoff = cpos16(mt, lg);
-      float p[4], ds[4];
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qloc = qoff + r;
-        const int qrow = qt + qloc;
-        float e = (scale * sa[r] - lse_lds[buf][qloc]) * LOG2E;
-        if (diag && (key > qrow || qrow >= S || key >= S)) e = -INFINITY;
-        p[r] = exp2f(e);
-        ds[r] = scale * p[r] * (da[r] - del_lds[buf][qloc]);
+      for (int half = 0; half < 2; ++half) {
+        const int mt = 2 * mth + half;
+        const f32x4 sa = half ? sa1 : sa0;
+        const f32x4 da = half ? da1 : da0;
+        const int qoff = cpos16(mt, lg);
+        float p[4], ds[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int qloc = qoff + r;
+          const int qrow = qt + qloc;
+          float e = (scale * sa[r] - lse_lds[buf][qloc]) * LOG2E;
+          if (diag && (key > qrow || qrow >= S || key >= S)) e = -INFINITY;
+          p[r] = exp2f(e);
+          ds[r] = scale * p[r] * (da[r] - del_lds[buf][qloc]);
+        }
+        const int kc = mt >> 1, rp = (mt & 1) * 2;
+        pk_p[kc][rp + 0] = cvt_pk_bf16(p[0], p[1]);
+        pk_p[kc][rp + 1] = cvt_pk_bf16(p[2], p[3]);
+        pk_ds[kc][rp + 0] = cvt_pk_bf16(ds[0], ds[1]);
+        pk_ds[kc][rp + 1] = cvt_pk_bf16(ds[2], ds[3]);
       }
-      const int kc = mt >> 1, rp = (mt & 1) * 2;
-      pk_p[kc][rp + 0] = cvt_pk_bf16(p[0], p[1]);
-      pk_p[kc][rp + 1] = cvt_pk_bf16(p[2], p[3]);
-      pk_ds[kc][rp + 0] = cvt_pk_bf16(ds[0], ds[1]);
-      pk_ds[kc][rp + 1] = cvt_pk_bf16(ds[2], ds[3]);
     }
 
     // ---- dV += P^T dO ; dK += dS^T Q (B operands via tr16 transpose
