@@ -8,6 +8,7 @@ required (gloo).
     torchrun --standalone --local-addr 127.0.0.1 --nproc-per-node 4 \
         --max-restarts 3 related-topics/elastic-training/toy.py
 """
+import datetime
 import json
 import os
 import random
@@ -23,7 +24,12 @@ STATE_FILE = os.environ.get("TOY_STATE_FILE", "toy-state.json")
 def main():
     rank = int(os.environ.get("RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))
-    dist.init_process_group("gloo")
+    # bounded collectives: when a peer dies, the survivors' barrier must
+    # throw promptly so torchelastic can tear down and restart the group
+    # (an unbounded gloo barrier holds the GIL in native code and stalls
+    # the SIGTERM teardown for minutes — see diagnosing-errors/README.md)
+    dist.init_process_group("gloo",
+                            timeout=datetime.timedelta(seconds=20))
     # NOTE: rank/world_size are NOT stable across restarts — reload shared
     # progress from the state file, never from process memory.
     state = {"iteration": 0}
