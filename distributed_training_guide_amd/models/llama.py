@@ -19,7 +19,6 @@ from dataclasses import dataclass, field
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from ..ops import (RMSNorm, causal_lm_loss, flash_attention, qkv_rope,
                    silu_mul)

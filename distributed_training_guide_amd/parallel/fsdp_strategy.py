@@ -7,7 +7,6 @@ import os
 from pathlib import Path
 
 import torch
-import torch.distributed as dist
 
 from ..models import build_model
 from ..ops import FusedAdamW

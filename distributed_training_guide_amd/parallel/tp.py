@@ -20,7 +20,6 @@ group by a post-accumulate-grad hook (DTensor gave the reference this for
 free; here it is explicit).
 """
 import logging
-import math
 
 import torch
 import torch.distributed as dist

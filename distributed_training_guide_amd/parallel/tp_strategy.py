@@ -4,9 +4,7 @@ import logging
 from pathlib import Path
 
 import torch
-import torch.distributed as dist
 
-from ..models import get_config
 from ..ops import FusedAdamW
 from ..trainer import pick_device
 from ..utils import checkpoint as ckpt

@@ -14,10 +14,8 @@ Extras the reference ships as diff-recipes are first-class flags:
 --wandb (wandb-configurations topologies).
 """
 import argparse
-import json
 import logging
 import os
-import time
 from contextlib import nullcontext
 from pathlib import Path
 

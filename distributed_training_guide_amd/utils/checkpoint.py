@@ -14,7 +14,6 @@ RNG state save/restore implements the determinism recipe
 (related-topics/determinism/README.md:46-68) behind save_rng/load_rng.
 """
 import json
-import os
 import random
 from pathlib import Path
 

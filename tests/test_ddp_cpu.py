@@ -1,8 +1,6 @@
 """Multi-process (gloo, world 2) tests of the DDP bucket engine and ZeRO-1
 — CPU stand-ins for the RCCL path per SURVEY.md §4."""
-import os
 
-import pytest
 import torch
 
 from utils_dist import run_dist

@@ -3,10 +3,8 @@ torch RNG state at checkpoints (the determinism recipe,
 reference related-topics/determinism/README.md:46-68), so 4+2 steps with a
 resume is BITWISE identical to 6 straight steps."""
 import importlib.util
-import json
 from pathlib import Path
 
-import pytest
 import torch
 
 REPO = Path(__file__).resolve().parent.parent

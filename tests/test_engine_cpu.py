@@ -2,7 +2,6 @@
 (distributed_training_guide_amd/engine.py) — the alternative-frameworks
 chapter's counterpart of deepspeed.initialize
 (/root/reference/alternative-frameworks/deepspeed/train_llm.py:58-73)."""
-import json
 
 import pytest
 import torch

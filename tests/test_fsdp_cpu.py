@@ -1,7 +1,6 @@
 """Multi-process (gloo) tests of the flat-param FSDP engine: grad parity
 with single-process training, sharded checkpoint round-trip + reshard,
 meta-init, activation checkpointing, no_sync accumulation."""
-import json
 
 import pytest
 import torch

@@ -2,7 +2,6 @@
 HIP kernels are compared against in test_ops_gpu.py)."""
 import math
 
-import pytest
 import torch
 import torch.nn.functional as F
 

@@ -2,7 +2,6 @@
 parity (SURVEY.md §7 — "off-by-one sharding bugs show up as silent loss
 divergence — needs layerwise activation-checksum tests against a TP=1
 run"), loss-parallel equivalence, chapter 6/7 end-to-end."""
-import pytest
 import torch
 
 from utils_dist import run_dist

@@ -1,10 +1,8 @@
 """download.py safetensors prefetch -> --weights-dir broadcast init
 (reference 05: download.py:6-21 + from_pretrained/broadcast 05:76-126)."""
 import importlib.util
-import sys
 from pathlib import Path
 
-import pytest
 import torch
 
 from utils_dist import run_dist

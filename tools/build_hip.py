@@ -12,7 +12,6 @@ newer.  Kernel files compile in parallel.
 import concurrent.futures as cf
 import os
 import subprocess
-import sys
 from pathlib import Path
 
 REPO = Path(__file__).resolve().parent.parent
