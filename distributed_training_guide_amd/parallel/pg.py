@@ -16,16 +16,26 @@ import torch
 import torch.distributed as dist
 
 
+def _env_int(*names, default):
+    for n in names:
+        v = os.environ.get(n)
+        if v is not None:
+            return int(v)
+    return default
+
+
 def env_rank() -> int:
-    return int(os.environ.get("RANK", 0))
+    # torchrun sets RANK; under mpirun OpenMPI sets OMPI_COMM_WORLD_RANK
+    # (the reference's mpi variant reads the same, 03-.../README.md:127-133)
+    return _env_int("RANK", "OMPI_COMM_WORLD_RANK", default=0)
 
 
 def env_world_size() -> int:
-    return int(os.environ.get("WORLD_SIZE", 1))
+    return _env_int("WORLD_SIZE", "OMPI_COMM_WORLD_SIZE", default=1)
 
 
 def env_local_rank() -> int:
-    return int(os.environ.get("LOCAL_RANK", 0))
+    return _env_int("LOCAL_RANK", "OMPI_COMM_WORLD_LOCAL_RANK", default=0)
 
 
 def init_distributed(device: torch.device | None = None,
@@ -35,6 +45,11 @@ def init_distributed(device: torch.device | None = None,
     rank, local_rank, world = env_rank(), env_local_rank(), env_world_size()
     if dist.is_initialized():
         return dist.get_rank(), local_rank, dist.get_world_size()
+    if "RANK" not in os.environ and "OMPI_COMM_WORLD_RANK" in os.environ:
+        # let torch's env:// rendezvous see the mpi-provided identity
+        os.environ["RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = str(world)
+        os.environ.setdefault("LOCAL_RANK", str(local_rank))
     if world == 1 and "MASTER_ADDR" not in os.environ:
         # single-process launch without torchrun: self-rendezvous
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
