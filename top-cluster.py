@@ -68,6 +68,8 @@ def main():
                    help="file listing hosts (one per line)")
     p.add_argument("--local", action="store_true", help="poll this node only")
     p.add_argument("--poll-freq", type=int, default=5000, help="ms")
+    p.add_argument("--iterations", type=int, default=0,
+                   help="stop after N polls (0 = run forever)")
     args = p.parse_args()
 
     if args.local or args.hosts is None:
@@ -77,6 +79,7 @@ def main():
         hosts = [h.strip() for h in open(args.hosts) if h.strip()]
         local = False
 
+    n_iter = 0
     while True:
         rows = []
         cl_util = cl_pow = cl_cap = cl_mem = cl_np = n_gpu = 0
@@ -108,6 +111,9 @@ def main():
             print(f"{'cluster':24s} {cl_util / n_gpu:6.1f}% "
                   f"{100 * cl_pow / max(cl_cap, 1):6.1f}% "
                   f"{cl_mem / n_gpu:6.1f}% {cl_np:>6d}")
+        n_iter += 1
+        if args.iterations and n_iter >= args.iterations:
+            break
         time.sleep(args.poll_freq / 1000.0)
 
 
