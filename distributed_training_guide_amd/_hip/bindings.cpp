@@ -162,6 +162,9 @@ std::vector<torch::Tensor> qkv_rope_fwd(torch::Tensor qkv, torch::Tensor cos,
   TORCH_CHECK(qkv.dim() == 3, "qkv_rope expects [B,S,W]");
   const int64_t B = qkv.size(0), S = qkv.size(1);
   TORCH_CHECK(qkv.size(2) == (Hq + 2 * Hkv) * D, "packed width mismatch");
+  TORCH_CHECK(D % 16 == 0,
+              "qkv_rope requires head_dim % 16 == 0 (8-pair vectorized "
+              "kernel); use the unfused rope path otherwise");
   const int* pos_ptr = nullptr;
   if (positions.has_value()) {
     TORCH_CHECK(positions->dtype() == torch::kInt &&
@@ -189,6 +192,7 @@ torch::Tensor qkv_rope_bwd(torch::Tensor dq, torch::Tensor dk,
   CHECK_BF16_CONTIG(dv);
   const int64_t B = dq.size(0), S = dq.size(1);
   const int64_t Hq = dq.size(2), Hkv = dk.size(2), D = dq.size(3);
+  TORCH_CHECK(D % 16 == 0, "qkv_rope requires head_dim % 16 == 0");
   const int* pos_ptr = nullptr;
   if (positions.has_value()) pos_ptr = positions->data_ptr<int>();
   auto dqkv = torch::empty({B, S, (Hq + 2 * Hkv) * D}, dq.options());

@@ -12,8 +12,9 @@ engine (parallel/fsdp.py — ZeRO stage 3 semantics: param + grad + optimizer
 state sharding with reduce-scatter/all-gather over RCCL/xGMI), the fused
 HIP AdamW, and the sharded checkpoint layout of utils/checkpoint.py.
 
-    engine, _, _, lr_scheduler = initialize(config="engine_config.json",
-                                            model_factory=lambda: model)
+    engine, _, _, lr_scheduler = initialize(
+        config="engine_config.json",
+        model_factory=lambda dtype: build_model(cfg, dtype=dtype))
     loss = engine(**batch).loss
     engine.backward(loss)
     engine.step()
@@ -173,9 +174,14 @@ def initialize(config, model_factory, device=None):
 
 
 def _cpu_optim_sd(optimizer):
+    # state_dict() returns REFERENCES to the live per-param state dicts;
+    # mutating them in place would swap the GPU moment tensors for CPU
+    # copies under the running FusedAdamW (whose descriptor cache holds raw
+    # device pointers).  Build a fresh nested dict instead.
     sd = optimizer.state_dict()
-    for st in sd["state"].values():
-        for k, v in st.items():
-            if isinstance(v, torch.Tensor):
-                st[k] = v.to("cpu")
+    sd["state"] = {
+        k: {kk: (vv.to("cpu") if isinstance(vv, torch.Tensor) else vv)
+            for kk, vv in st.items()}
+        for k, st in sd["state"].items()
+    }
     return sd

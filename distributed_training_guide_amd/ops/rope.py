@@ -24,6 +24,21 @@ def get_rope_tables(dim: int, max_pos: int, theta: float, device):
     return entry
 
 
+def _table_size(S: int, positions, max_pos) -> int:
+    """Table rows needed.  When the caller supplies max_pos (the
+    config-known max position, as models/llama.py does) we trust it and
+    avoid a GPU->CPU `positions.max().item()` sync per attention layer;
+    only positions on CPU or with no declared bound pay the reduction."""
+    need = max_pos or S
+    if positions is None:
+        return need
+    if max_pos is not None and not positions.is_cuda:
+        return max(need, int(positions.max()) + 1)
+    if max_pos is None:
+        return max(need, int(positions.max().item()) + 1)
+    return need
+
+
 class _RoPEFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, cos, sin, positions):
@@ -64,14 +79,22 @@ def qkv_rope(qkv: torch.Tensor, Hq: int, Hkv: int, D: int,
     """Split packed qkv [B,S,(Hq+2Hkv)*D] and apply RoPE to q/k in one
     fused pass. Returns (q [B,S,Hq,D], k, v [B,S,Hkv,D])."""
     B, S, W = qkv.shape
-    need = max_pos or S
+    need = _table_size(S, positions, max_pos)
+    cos, sin = get_rope_tables(D, need, theta, qkv.device)
     if positions is not None:
         positions = positions.to(torch.int32).contiguous()
-        need = max(need, int(positions.max().item()) + 1)
-    cos, sin = get_rope_tables(D, need, theta, qkv.device)
-    if use_hip(qkv):
+    if use_hip(qkv) and D % 16 == 0:
         return _QKVRopeFn.apply(qkv.contiguous(), cos, sin, positions,
                                 Hq, Hkv, D)
+    if use_hip(qkv):
+        # head_dim not 16-aligned: the fused kernel's 8-pair vectorization
+        # doesn't apply — split + per-tensor HIP rope instead.
+        q, k, v = qkv.split([Hq * D, Hkv * D, Hkv * D], dim=-1)
+        q = _RoPEFn.apply(q.reshape(B, S, Hq, D).contiguous(), cos, sin,
+                          positions)
+        k = _RoPEFn.apply(k.reshape(B, S, Hkv, D).contiguous(), cos, sin,
+                          positions)
+        return q, k, v.reshape(B, S, Hkv, D).contiguous()
     # CPU fallback: split views + eager rope
     q, k, v = qkv.split([Hq * D, Hkv * D, Hkv * D], dim=-1)
     q = rope_ref(q.view(B, S, Hq, D), cos, sin, positions)
@@ -84,11 +107,10 @@ def rope(x: torch.Tensor, theta: float = 10000.0,
          max_pos: int | None = None) -> torch.Tensor:
     """Apply rotate-half RoPE to x [B, S, H, D]."""
     B, S, H, D = x.shape
-    need = max_pos or S
+    need = _table_size(S, positions, max_pos)
+    cos, sin = get_rope_tables(D, need, theta, x.device)
     if positions is not None:
         positions = positions.to(torch.int32).contiguous()
-        need = max(need, int(positions.max().item()) + 1)
-    cos, sin = get_rope_tables(D, need, theta, x.device)
     if use_hip(x):
         return _RoPEFn.apply(x.contiguous(), cos, sin, positions)
     return rope_ref(x, cos, sin, positions)
