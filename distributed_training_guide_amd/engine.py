@@ -134,7 +134,7 @@ class Engine(torch.nn.Module):
         state = dict(client_state or {})
         state["global_step"] = self.global_step
         ckpt.save_sharded(exp_dir, self.module.sharded_state_dict(),
-                          _cpu_optim_sd(self.optimizer), self.lr_scheduler,
+                          ckpt.optim_sd_cpu(self.optimizer), self.lr_scheduler,
                           state, self.rank, self.world_size)
 
     def load_checkpoint(self, save_dir):
@@ -171,17 +171,3 @@ def initialize(config, model_factory, device=None):
         model = model_factory(dtype)
     engine = Engine(model, config, device)
     return engine, engine.optimizer, None, engine.lr_scheduler
-
-
-def _cpu_optim_sd(optimizer):
-    # state_dict() returns REFERENCES to the live per-param state dicts;
-    # mutating them in place would swap the GPU moment tensors for CPU
-    # copies under the running FusedAdamW (whose descriptor cache holds raw
-    # device pointers).  Build a fresh nested dict instead.
-    sd = optimizer.state_dict()
-    sd["state"] = {
-        k: {kk: (vv.to("cpu") if isinstance(vv, torch.Tensor) else vv)
-            for kk, vv in st.items()}
-        for k, st in sd["state"].items()
-    }
-    return sd

@@ -73,7 +73,7 @@ class FSDPStrategy:
     def save_checkpoint(self, exp_dir: Path, model, optimizer, lr_scheduler,
                         state):
         ckpt.save_sharded(exp_dir, model.sharded_state_dict(),
-                          _optim_sd_cpu(optimizer), lr_scheduler, state,
+                          ckpt.optim_sd_cpu(optimizer), lr_scheduler, state,
                           self.rank, self.world_size)
 
     def load_checkpoint(self, exp_dir: Path, model, optimizer, lr_scheduler):
@@ -90,15 +90,6 @@ class FSDPStrategy:
                               map_location="cpu", weights_only=True)
         lr_scheduler.load_state_dict(sched_sd)
         return state
-
-
-def _optim_sd_cpu(optimizer):
-    sd = optimizer.state_dict()
-    for st in sd["state"].values():
-        for k, v in st.items():
-            if isinstance(v, torch.Tensor):
-                st[k] = v.to("cpu")
-    return sd
 
 
 def _load_resharding(exp_dir: Path, model, optimizer, rank):

@@ -58,7 +58,7 @@ class TPStrategy:
         # DCP-style: every rank writes its tp shard (06:171-174)
         inner = model.module if hasattr(model, "module") else model
         ckpt.save_sharded(exp_dir, inner.tp_state_dict(),
-                          _cpu_sd(optimizer.state_dict()), lr_scheduler,
+                          ckpt.optim_sd_cpu(optimizer), lr_scheduler,
                           state, self.rank, self.world_size)
 
     def load_checkpoint(self, exp_dir: Path, model, optimizer, lr_scheduler):
@@ -103,7 +103,7 @@ class TwoDStrategy(TPStrategy):
         # shard-of-shard: each (dp, tp) rank saves its FSDP shard of the
         # local tp shard (the reference's 2D DCP save, 07:172-175)
         ckpt.save_sharded(exp_dir, model.sharded_state_dict(),
-                          _cpu_sd(optimizer.state_dict()), lr_scheduler,
+                          ckpt.optim_sd_cpu(optimizer), lr_scheduler,
                           state, self.rank, self.world_size)
 
     def load_checkpoint(self, exp_dir: Path, model, optimizer, lr_scheduler):
@@ -115,11 +115,3 @@ class TwoDStrategy(TPStrategy):
                               map_location="cpu", weights_only=True)
         lr_scheduler.load_state_dict(sched_sd)
         return state
-
-
-def _cpu_sd(sd):
-    for st in sd.get("state", {}).values():
-        for k, v in st.items():
-            if isinstance(v, torch.Tensor):
-                st[k] = v.to("cpu")
-    return sd

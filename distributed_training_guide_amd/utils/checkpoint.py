@@ -27,6 +27,22 @@ def _barrier():
         dist.barrier()
 
 
+def optim_sd_cpu(optimizer) -> dict:
+    """Optimizer state dict with all tensors copied to CPU, WITHOUT
+    mutating the live state.  torch's state_dict() returns references to
+    the live per-param state dicts: assigning `st[k] = v.to('cpu')` in
+    place would swap the GPU moment buffers for CPU copies under a running
+    FusedAdamW (whose chunk-descriptor cache holds raw device pointers) —
+    every later fused step would write through dangling pointers."""
+    sd = optimizer.state_dict()
+    sd["state"] = {
+        k: {kk: (vv.to("cpu") if isinstance(vv, torch.Tensor) else vv)
+            for kk, vv in st.items()}
+        for k, st in sd["state"].items()
+    }
+    return sd
+
+
 def mkdir_rank0(path: Path, rank: int):
     """Mount-aware mkdir under barrier fencing (reference 04:160-168)."""
     _barrier()
