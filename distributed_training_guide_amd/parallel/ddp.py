@@ -21,6 +21,7 @@ saturation while still giving several overlap chunks per backward
 (--bucket-cap-mb tunable; the reference's 500 MB was an NVLink choice).
 """
 import logging
+import os
 from contextlib import contextmanager
 
 import torch
@@ -28,6 +29,13 @@ import torch.distributed as dist
 from torch import nn
 
 LOGGER = logging.getLogger(__name__)
+
+
+def _force_collectives():
+    """DTGA_FORCE_COLLECTIVES=1: run the real RCCL collective branch even
+    at world=1 (see parallel/fsdp.py) so 1-GPU runs execute the exact
+    multi-GPU call pattern."""
+    return os.environ.get("DTGA_FORCE_COLLECTIVES") == "1"
 
 
 class _Bucket:
@@ -98,7 +106,7 @@ class DistributedDataParallel(nn.Module):
 
     # ---- construction-time state broadcast (reference README 02:177) ----
     def _broadcast_module_states(self):
-        if self.world_size == 1:
+        if self.world_size == 1 and not _force_collectives():
             return
         with torch.no_grad():
             for t in list(self.module.parameters()) + \
@@ -125,7 +133,8 @@ class DistributedDataParallel(nn.Module):
                 # fold it into the bucket view and re-alias.
                 b.views[param].copy_(param.grad)
                 param.grad = b.views[param]
-            if not self.sync_enabled or self.world_size == 1:
+            if not self.sync_enabled or (self.world_size == 1
+                                         and not _force_collectives()):
                 return
             if not self._final_cb_armed:
                 self._final_cb_armed = True

@@ -31,6 +31,7 @@ grads and reshards.  An end-of-backward engine callback waits the in-flight
 reduce-scatters and accumulates into the shard gradients.
 """
 import logging
+import os
 from contextlib import contextmanager
 
 import torch
@@ -38,6 +39,14 @@ import torch.distributed as dist
 from torch import nn
 
 LOGGER = logging.getLogger(__name__)
+
+
+def _force_collectives():
+    """DTGA_FORCE_COLLECTIVES=1: execute the real RCCL collective branch
+    even at world=1 (single-rank communicators are valid; collectives
+    degenerate to device copies) so a 1-GPU box can run and profile the
+    exact multi-GPU call pattern."""
+    return os.environ.get("DTGA_FORCE_COLLECTIVES") == "1"
 
 
 def _pad(n, m):
@@ -157,7 +166,8 @@ class _FSDPUnit:
         src = self.shard.data
         if self.cpu_offload:
             src = src.to(self.device, non_blocking=True)
-        if self.world == 1:
+        if self.world == 1 and not (_force_collectives()
+                                    and self.flat.is_cuda):
             self.flat.copy_(src)
             self.is_unsharded = True
             return
@@ -200,7 +210,8 @@ class _FSDPUnit:
         flat_g.div_(self.world)
         out = torch.empty(self.shard_numel, dtype=self.reduce_dtype,
                           device=flat_g.device)
-        if self.world == 1:
+        if self.world == 1 and not (_force_collectives()
+                                    and flat_g.is_cuda):
             out.copy_(flat_g)
             self._rs_work, self._rs_out = None, out
         elif flat_g.is_cuda:

@@ -107,10 +107,17 @@ class _GatherVocab(torch.autograd.Function):
         ctx.group = group
         world = dist.get_world_size(group)
         ctx.vlocal = x.shape[-1]
-        parts = [torch.empty_like(x.contiguous()) for _ in range(world)]
-        dist.all_gather(parts, x.contiguous(), group=group)
-        parts[dist.get_rank(group)] = x  # keep autograd storage identity
-        return torch.cat(parts, dim=-1)
+        xc = x.contiguous()
+        parts = torch.empty((world, *xc.shape), dtype=x.dtype,
+                            device=x.device)
+        if x.is_cuda:
+            dist.all_gather_into_tensor(parts, xc, group=group)
+        else:
+            dist.all_gather(list(parts.unbind(0)), xc, group=group)
+        # concat rank chunks on the vocab (last) dim: one permute copy
+        perm = (*range(1, x.dim()), 0, x.dim())
+        return parts.permute(perm).reshape(*x.shape[:-1],
+                                           world * ctx.vlocal)
 
     @staticmethod
     def backward(ctx, g):
@@ -118,20 +125,30 @@ class _GatherVocab(torch.autograd.Function):
         return g[..., r * ctx.vlocal: (r + 1) * ctx.vlocal].contiguous(), None
 
 
+# DTGA_FORCE_COLLECTIVES=1 runs the real collective branch even at
+# world=1 (RCCL supports single-rank communicators: collectives become
+# device copies) — lets a 1-GPU box execute and profile the exact RCCL
+# call pattern of the multi-GPU path.
+def _force_collectives():
+    import os
+
+    return os.environ.get("DTGA_FORCE_COLLECTIVES") == "1"
+
+
 def gather_seq(x, group):
-    if dist.get_world_size(group) == 1:
+    if dist.get_world_size(group) == 1 and not _force_collectives():
         return x
     return _GatherSeq.apply(x, group)
 
 
 def reduce_scatter_seq(x, group):
-    if dist.get_world_size(group) == 1:
+    if dist.get_world_size(group) == 1 and not _force_collectives():
         return x
     return _ReduceScatterSeq.apply(x, group)
 
 
 def gather_vocab(x, group):
-    if dist.get_world_size(group) == 1:
+    if dist.get_world_size(group) == 1 and not _force_collectives():
         return x
     return _GatherVocab.apply(x, group)
 
