@@ -22,6 +22,7 @@ import torch.nn as nn
 
 from ..ops import (RMSNorm, causal_lm_loss, flash_attention, qkv_rope,
                    silu_mul)
+from ..ops.fused_linear_ce import fused_causal_lm_loss
 from ..ops.rmsnorm import add_rmsnorm
 
 
@@ -212,6 +213,13 @@ class LlamaForCausalLM(nn.Module):
             for layer in self.layers:
                 x = layer(x, position_ids)
             x = self.norm(x)
+        if labels is not None and x.is_cuda:
+            # chunked fused projection+CE: the [B,S,V] logits (+grad) are
+            # never materialized (ops/fused_linear_ce.py) — logits=None,
+            # like the reference's loss-only training consumers (02:149-163
+            # only reads outputs.loss)
+            loss = fused_causal_lm_loss(x, self.lm_head.weight, labels)
+            return CausalLMOutput(loss=loss, logits=None)
         logits = self.lm_head(x)
         loss = None
         if labels is not None:

@@ -80,6 +80,8 @@ class _FSDPUnit:
 
         # full flat tensor; its storage is resized 0<->full
         self.flat = torch.empty(self.padded, dtype=self.dtype, device=device)
+        self.alias_shard = (self.world == 1 and not cpu_offload
+                            and not _force_collectives())
         self.was_meta = any(p.is_meta for p in self.params)
         if not self.was_meta:
             self._materialize_initial()
@@ -125,6 +127,21 @@ class _FSDPUnit:
         self.sync_enabled = True
 
     def create_shard(self):
+        if self.alias_shard:
+            # world=1, no offload: the "shard" IS the flat param (identical
+            # size) — alias it instead of keeping a duplicate copy (saves a
+            # full model copy, 16 GB for 8B bf16) and make shard/unshard
+            # no-ops.  .data assignment keeps the Parameter from becoming
+            # an autograd view of flat.
+            self.shard = nn.Parameter(
+                torch.empty(0, dtype=self.dtype, device=self.device))
+            self.shard.data = self.flat
+            if self.reduce_dtype != self.dtype:
+                try:
+                    self.shard.grad_dtype = self.reduce_dtype
+                except (AttributeError, RuntimeError):
+                    pass
+            return
         shard_dev = torch.device("cpu") if self.cpu_offload else self.device
         shard_src = self.flat[self.rank * self.shard_numel:
                               (self.rank + 1) * self.shard_numel]
@@ -152,6 +169,8 @@ class _FSDPUnit:
 
     # ---- shard/unshard ----
     def reshard(self):
+        if self.alias_shard:
+            return  # shard aliases flat: nothing to free
         if not self.is_unsharded:
             return
         self.flat.untyped_storage().resize_(0)
@@ -194,20 +213,30 @@ class _FSDPUnit:
 
     # ---- gradient reduce-scatter ----
     def reduce_scatter_grads(self):
-        grads = []
-        for p, numel in zip(self.params, self.numels):
-            g = p.grad
-            if g is None:
-                g = torch.zeros(numel, dtype=self.reduce_dtype,
-                                device=self.flat.device)
-            grads.append(g.reshape(-1).to(self.reduce_dtype))
-        flat_g = torch.cat(grads)
-        if flat_g.numel() < self.padded:
-            flat_g = torch.cat([
-                flat_g,
-                torch.zeros(self.padded - flat_g.numel(),
-                            dtype=self.reduce_dtype, device=flat_g.device)])
-        flat_g.div_(self.world)
+        # pack grads into ONE flat reduce-dtype buffer: a single cast-copy
+        # per param straight into its slice (no fp32 intermediates, no
+        # torch.cat), then one fp32 pre-division pass (grad/world THEN sum
+        # — the reference's stated DDP semantics, 02-.../README.md:179-185;
+        # dividing after the reduce instead changes Adam's step on
+        # near-cancelling elements).  The buffer is a fresh same-sized
+        # alloc each backward, so the caching allocator reuses one block
+        # per unit while the async RS is in flight.
+        dev = self.flat.device
+        flat_g = torch.empty(self.padded, dtype=self.reduce_dtype,
+                             device=dev)
+        off = 0
+        with torch.no_grad():
+            for p, numel in zip(self.params, self.numels):
+                dst = flat_g[off: off + numel]
+                if p.grad is None:
+                    dst.zero_()
+                else:
+                    dst.copy_(p.grad.reshape(-1))
+                off += numel
+            if off < self.padded:
+                flat_g[off:].zero_()
+            if self.world > 1:
+                flat_g.div_(self.world)
         out = torch.empty(self.shard_numel, dtype=self.reduce_dtype,
                           device=flat_g.device)
         if self.world == 1 and not (_force_collectives()
@@ -360,8 +389,9 @@ class FSDP(nn.Module):
                     for n, v in zip(u.param_names, u.views):
                         init_fn(n, v)
                 u.create_shard()
-                u.flat.untyped_storage().resize_(0)
-                u.is_unsharded = False
+                if not u.alias_shard:
+                    u.flat.untyped_storage().resize_(0)
+                    u.is_unsharded = False
             else:
                 u.create_shard()
 

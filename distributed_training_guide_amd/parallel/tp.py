@@ -34,39 +34,43 @@ LOGGER = logging.getLogger(__name__)
 
 
 # ---------------------------------------------------------------- collectives
-def _seq_to_front(x):
-    # [B, S, H] -> [S, B, H] contiguous (collectives shard dim 0)
-    return x.transpose(0, 1).contiguous()
+#
+# Layout strategy (MI355X: every copy is a full-activation HBM round trip,
+# 4 boundaries x n_layers x fwd+bwd of them per step): the seq-sharded
+# [B, s, H] activation is ALREADY a contiguous rank chunk of the
+# [tp, B, s, H] gathered stack, so the all-gather needs NO input copy —
+# only one permute copy to interleave ranks back into [B, tp*s, H].  The
+# reduce-scatter is its mirror: one permute copy into [tp, B, s, H] rank
+# order, and the collective writes the output directly in final layout.
+# One copy per boundary, versus two with the [B,S,H]<->[S,B,H] transpose
+# pair this replaces.
 
-
-def _seq_to_back(x):
-    return x.transpose(0, 1).contiguous()
-
-
-def _all_gather_dim0(x, group):
+def _gather_seq_1copy(x, group):
+    """[B, s, H] -> [B, tp*s, H] with one permute copy after the gather."""
     world = dist.get_world_size(group)
-    out = torch.empty((x.shape[0] * world, *x.shape[1:]), dtype=x.dtype,
-                      device=x.device)
+    xc = x.contiguous()
+    B, s, H = xc.shape
+    parts = torch.empty((world, B, s, H), dtype=x.dtype, device=x.device)
     if x.is_cuda:
-        dist.all_gather_into_tensor(out, x, group=group)
+        dist.all_gather_into_tensor(parts, xc, group=group)
     else:
-        chunks = list(out.chunk(world))
-        dist.all_gather(chunks, x.contiguous(), group=group)
-    return out
+        dist.all_gather(list(parts.unbind(0)), xc, group=group)
+    return parts.permute(1, 0, 2, 3).reshape(B, world * s, H)
 
 
-def _reduce_scatter_dim0(x, group):
+def _reduce_scatter_seq_1copy(x, group):
+    """[B, S, H] partial sums -> [B, S/tp, H] with one permute copy before
+    the reduce-scatter; the output lands directly in its final layout."""
     world = dist.get_world_size(group)
-    out = torch.empty((x.shape[0] // world, *x.shape[1:]), dtype=x.dtype,
-                      device=x.device)
+    B, S, H = x.shape
+    s = S // world
+    inp = x.reshape(B, world, s, H).permute(1, 0, 2, 3).contiguous()
+    out = torch.empty((B, s, H), dtype=x.dtype, device=x.device)
     if x.is_cuda:
-        dist.reduce_scatter_tensor(out, x.contiguous(), group=group)
-    else:  # gloo: all-reduce then slice
-        y = x.contiguous().clone()
-        dist.all_reduce(y, group=group)
-        r = dist.get_rank(group)
-        n = out.shape[0]
-        out.copy_(y[r * n: (r + 1) * n])
+        dist.reduce_scatter_tensor(out, inp, group=group)
+    else:  # gloo: all-reduce then take this rank's chunk
+        dist.all_reduce(inp, group=group)
+        out.copy_(inp[dist.get_rank(group)])
     return out
 
 
@@ -76,12 +80,11 @@ class _GatherSeq(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, group):
         ctx.group = group
-        return _seq_to_back(_all_gather_dim0(_seq_to_front(x), group))
+        return _gather_seq_1copy(x, group)
 
     @staticmethod
     def backward(ctx, g):
-        return _seq_to_back(
-            _reduce_scatter_dim0(_seq_to_front(g), ctx.group)), None
+        return _reduce_scatter_seq_1copy(g, ctx.group), None
 
 
 class _ReduceScatterSeq(torch.autograd.Function):
@@ -90,12 +93,11 @@ class _ReduceScatterSeq(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, group):
         ctx.group = group
-        return _seq_to_back(_reduce_scatter_dim0(_seq_to_front(x), group))
+        return _reduce_scatter_seq_1copy(x, group)
 
     @staticmethod
     def backward(ctx, g):
-        return _seq_to_back(
-            _all_gather_dim0(_seq_to_front(g), ctx.group)), None
+        return _gather_seq_1copy(g, ctx.group), None
 
 
 class _GatherVocab(torch.autograd.Function):
