@@ -86,7 +86,7 @@ class _FusedLinearCEFn(torch.autograd.Function):
 
 def fused_causal_lm_loss(x: torch.Tensor, weight: torch.Tensor,
                          labels: torch.Tensor,
-                         chunk_rows: int = 4096) -> torch.Tensor:
+                         chunk_rows: int = 8192) -> torch.Tensor:
     """Mean causal-LM CE of linear(x, weight) against shifted labels,
     without materializing [B,S,V] logits.  x [B,S,H] bf16 contiguous,
     weight [V,H] (V padded to a multiple of 8), labels [B,S] int64."""

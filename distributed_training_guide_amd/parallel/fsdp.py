@@ -338,23 +338,30 @@ class FSDP(nn.Module):
             layer_cls = (LlamaDecoderLayer, GPT2Block)
 
         # ---- partition params into units ----
+        # Coverage is tracked by QUALIFIED NAME, not object identity:
+        # building a unit from a meta-built model REPLACES its Parameters
+        # with fresh objects (see _FSDPUnit), so an identity set would
+        # never match module.named_parameters() afterwards and every
+        # layer param would be double-assigned to the root unit (one
+        # all-params root unit + dead per-layer units).
         name_of = {p: n for n, p in module.named_parameters()}
         layer_modules = [(n, m) for n, m in module.named_modules()
                          if isinstance(m, layer_cls)]
-        covered = set()
+        covered_names = set()
         self.units: list[_FSDPUnit] = []
         self._unit_of_module = {}
         for n, m in layer_modules:
             nps = [(name_of[p], p) for p in m.parameters()
                    if p.requires_grad]
-            covered.update(p for _, p in nps)
+            covered_names.update(pn for pn, _ in nps)
             u = _FSDPUnit(n, m, nps, process_group, reshard_after_forward,
                           reduce_dtype, cpu_offload, device,
                           root_module=module)
             self.units.append(u)
             self._unit_of_module[m] = u
+        # re-read named_parameters: layer units may have swapped objects
         root_nps = [(n, p) for n, p in module.named_parameters()
-                    if p.requires_grad and p not in covered]
+                    if p.requires_grad and n not in covered_names]
         self.root_unit = None
         if root_nps:
             self.root_unit = _FSDPUnit("__root__", module, root_nps,

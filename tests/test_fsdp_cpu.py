@@ -90,6 +90,51 @@ def test_fsdp_meta_init():
     run_dist(_fsdp_meta_init, world_size=2)
 
 
+def test_meta_init_unit_partition_no_duplication():
+    """Regression: meta-built units replace Parameter objects, which made
+    the identity-based root partition re-cover every layer param — the
+    root unit silently became an all-params unit (whole model duplicated,
+    dead per-layer units).  Units must tile the param set exactly."""
+    import os
+
+    import torch.distributed as dist
+
+    from distributed_training_guide_amd.models import build_model
+    from distributed_training_guide_amd.parallel.fsdp import FSDP
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29529")
+    created = False
+    if not dist.is_initialized():
+        dist.init_process_group("gloo", rank=0, world_size=1)
+        created = True
+    try:
+        with torch.device("meta"):
+            model = build_model("llama-debug")
+        n_layers = len(model.layers)
+        total = sum(p.numel() for p in model.parameters())
+        fsdp = FSDP(model, device=torch.device("cpu"))
+        assert len(fsdp.units) == n_layers + 1
+        assert sum(u.total for u in fsdp.units) == total
+        per_layer = sum(p.numel()
+                        for p in model.layers[0].parameters())
+        assert fsdp.root_unit.total == total - n_layers * per_layer
+        # module params must be views of their unit's flat storage
+        name_to_unit = {}
+        for u in fsdp.units:
+            for pn in u.param_names:
+                assert pn not in name_to_unit, f"{pn} in two units"
+                name_to_unit[pn] = u
+        for n, p in model.named_parameters():
+            u = name_to_unit[n]
+            assert p.data_ptr() >= u.flat.data_ptr()
+            assert p.data_ptr() < u.flat.data_ptr() + \
+                u.flat.numel() * u.flat.element_size(), n
+    finally:
+        if created:
+            dist.destroy_process_group()
+
+
 def _fsdp_ckpt_roundtrip(rank, world, tmpdir):
     from pathlib import Path
 
