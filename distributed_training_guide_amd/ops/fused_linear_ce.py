@@ -3,12 +3,18 @@
 The reference's biggest single activation is the [B,S,V] logits tensor
 (/root/reference/06-tensor-parallel/README.md:243-271 motivates
 loss-parallel for exactly this); at Llama-3 vocab (128256) and bs24/s1024
-that is ~6.3 GB bf16 forward + ~6.3 GB gradient.  Here the final
-projection and the CE loss are computed in row chunks: each chunk's logits
-live only long enough to feed the online-LSE CE kernel (forward) or the
-softmax-grad + two GEMMs (backward, logits recomputed), so the full logits
-tensor is NEVER materialized.  Costs one extra lm_head GEMM in backward
-(~2%% of step at bs24); saves ~12 GB peak at bs24.
+that is ~6.3 GB bf16 forward + ~6.3 GB gradient.  Two fused modes:
+
+* keep_logits=True ("semi"): one full forward GEMM whose logits are saved,
+  backward chunks the softmax-grad + dgrad/wgrad GEMMs so the full
+  [B,S,V] GRADIENT never exists.  ~free; saves ~6.3 GB at bs24.
+* keep_logits=False ("fused"): forward also runs in chunks and saves
+  nothing; backward recomputes each logits chunk.  Costs one extra
+  lm_head GEMM (~2.5%% of step at bs24); saves ~12.6 GB.
+
+DTGA_CE_MODE selects: auto (default — fused only when free HBM is tight),
+semi, fused, unfused (full logits + plain CE, the fastest when memory is
+plentiful).
 
 The causal shift is folded into a precomputed flat shifted-label vector
 (position s predicts labels[b,s+1]; the last position of every row becomes
@@ -16,8 +22,10 @@ ignore_index), which lets chunks be plain contiguous row ranges of the
 flattened [B*S, H] input — chunk GEMMs need no copies and batch-row
 boundaries need no special casing.
 
-dW is accumulated across chunks in fp32 and cast to the weight dtype once.
+Multi-chunk dW is accumulated in fp32 and cast to the weight dtype once.
 """
+import os
+
 import torch
 
 from .._ext import ext
@@ -39,60 +47,104 @@ def _shifted_flat_labels(labels: torch.Tensor) -> torch.Tensor:
 
 class _FusedLinearCEFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, labels, chunk_rows):
+    def forward(ctx, x, weight, labels, chunk_rows, keep_logits):
         B, S, H = x.shape
-        V = weight.shape[0]
         x2 = x.reshape(B * S, H)
         N = B * S
         shifted = _shifted_flat_labels(labels)
         n_valid = max(int((shifted != IGNORE_INDEX).sum().item()), 1)
-        lse = torch.empty(N, dtype=torch.float32, device=x.device)
-        loss_sum = torch.zeros((), dtype=torch.float32, device=x.device)
-        wt = weight.t()
-        for r0 in range(0, N, chunk_rows):
-            r1 = min(r0 + chunk_rows, N)
-            logits_c = torch.matmul(x2[r0:r1], wt)  # [L,V] bf16, transient
-            lc, lse_c = ext().ce_fwd(logits_c.unsqueeze(0), shifted[r0:r1],
-                                     r1 - r0, IGNORE_INDEX)
-            loss_sum += lc.sum()
-            lse[r0:r1] = lse_c
-        ctx.save_for_backward(x2, weight, shifted, lse)
+        logits = None
+        if keep_logits:
+            logits = torch.matmul(x2, weight.t())  # [N,V], saved
+            loss_rows, lse = ext().ce_fwd(logits.unsqueeze(0), shifted, N,
+                                          IGNORE_INDEX)
+            loss_sum = loss_rows.sum()
+        else:
+            lse = torch.empty(N, dtype=torch.float32, device=x.device)
+            loss_sum = torch.zeros((), dtype=torch.float32, device=x.device)
+            wt = weight.t()
+            for r0 in range(0, N, chunk_rows):
+                r1 = min(r0 + chunk_rows, N)
+                logits_c = torch.matmul(x2[r0:r1], wt)  # transient
+                lc, lse_c = ext().ce_fwd(logits_c.unsqueeze(0),
+                                         shifted[r0:r1], r1 - r0,
+                                         IGNORE_INDEX)
+                loss_sum += lc.sum()
+                lse[r0:r1] = lse_c
+        if keep_logits:
+            ctx.save_for_backward(x2, weight, shifted, lse, logits)
+        else:
+            ctx.save_for_backward(x2, weight, shifted, lse)
         ctx.n_valid = n_valid
         ctx.chunk_rows = chunk_rows
         ctx.in_shape = (B, S, H)
+        ctx.keep_logits = keep_logits
         return loss_sum / n_valid
 
     @staticmethod
     def backward(ctx, dloss):
-        x2, weight, shifted, lse = ctx.saved_tensors
+        if ctx.keep_logits:
+            x2, weight, shifted, lse, logits = ctx.saved_tensors
+        else:
+            x2, weight, shifted, lse = ctx.saved_tensors
+            logits = None
         B, S, H = ctx.in_shape
         V = weight.shape[0]
         N = B * S
         scale = float(dloss.item()) / ctx.n_valid
         dx2 = torch.empty_like(x2)
-        dw32 = torch.zeros(V, H, dtype=torch.float32, device=x2.device)
+        one_chunk = ctx.chunk_rows >= N
+        # multi-chunk: accumulate dW in fp32 across chunks; single chunk:
+        # one bf16 GEMM, exactly the unfused wgrad
+        dw32 = None if one_chunk else torch.zeros(
+            V, H, dtype=torch.float32, device=x2.device)
+        dw = None
         wt = weight.t()
         for r0 in range(0, N, ctx.chunk_rows):
             r1 = min(r0 + ctx.chunk_rows, N)
             L = r1 - r0
-            logits_c = torch.matmul(x2[r0:r1], wt)  # recompute, transient
+            logits_c = (logits[r0:r1] if logits is not None
+                        else torch.matmul(x2[r0:r1], wt))  # recompute
             dlogits_c = ext().ce_bwd(logits_c.unsqueeze(0), shifted[r0:r1],
                                      lse[r0:r1], scale, L, 0, IGNORE_INDEX,
                                      False).squeeze(0)
+            del logits_c
             torch.matmul(dlogits_c, weight, out=dx2[r0:r1])
-            dw32.add_(torch.matmul(dlogits_c.t(), x2[r0:r1]))
-        return (dx2.view(B, S, H), dw32.to(weight.dtype), None, None)
+            if one_chunk:
+                dw = torch.matmul(dlogits_c.t(), x2[r0:r1])
+            else:
+                dw32.add_(torch.matmul(dlogits_c.t(), x2[r0:r1]))
+        if dw is None:
+            dw = dw32.to(weight.dtype)
+        return (dx2.view(B, S, H), dw, None, None, None)
 
 
 def fused_causal_lm_loss(x: torch.Tensor, weight: torch.Tensor,
                          labels: torch.Tensor,
-                         chunk_rows: int = 8192) -> torch.Tensor:
-    """Mean causal-LM CE of linear(x, weight) against shifted labels,
-    without materializing [B,S,V] logits.  x [B,S,H] bf16 contiguous,
-    weight [V,H] (V padded to a multiple of 8), labels [B,S] int64."""
-    if x.is_cuda:
-        return _FusedLinearCEFn.apply(x.contiguous(), weight,
-                                      labels.contiguous(), chunk_rows)
-    # CPU fallback: full logits + eager reference (tests; memory is moot)
-    logits = torch.matmul(x, weight.t())
-    return cross_entropy_ref(logits, labels, IGNORE_INDEX)
+                         chunk_rows: int | None = None) -> torch.Tensor:
+    """Mean causal-LM CE of linear(x, weight) against shifted labels
+    without materializing the full [B,S,V] logits gradient (and, under
+    memory pressure, without the logits themselves).  x [B,S,H] bf16
+    contiguous, weight [V,H] (V padded to a multiple of 8), labels [B,S]
+    int64.  DTGA_CE_MODE / DTGA_CE_CHUNK override the policy."""
+    if not x.is_cuda:
+        # CPU fallback: full logits + eager reference (tests; memory moot)
+        logits = torch.matmul(x, weight.t())
+        return cross_entropy_ref(logits, labels, IGNORE_INDEX)
+    mode = os.environ.get("DTGA_CE_MODE", "auto")
+    if mode == "unfused":
+        from .cross_entropy import causal_lm_loss
+
+        return causal_lm_loss(torch.matmul(x, weight.t()), labels)
+    if chunk_rows is None:
+        chunk_rows = int(os.environ.get("DTGA_CE_CHUNK", "8192"))
+    if mode == "semi":
+        keep = True
+    elif mode == "fused":
+        keep = False
+    else:  # auto: recompute-in-backward only when HBM is actually tight
+        logits_bytes = x.numel() // x.shape[-1] * weight.shape[0] * 2
+        free, _ = torch.cuda.mem_get_info(x.device)
+        keep = free > 4 * logits_bytes
+    return _FusedLinearCEFn.apply(x.contiguous(), weight,
+                                  labels.contiguous(), chunk_rows, keep)

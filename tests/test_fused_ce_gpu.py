@@ -15,12 +15,18 @@ def _ref_fp32(x, w, labels):
     return torch.nn.functional.cross_entropy(lg, lb, ignore_index=-100)
 
 
+@pytest.mark.parametrize("mode", ["fused", "semi"])
 @pytest.mark.parametrize("B,S,H,V,chunk", [
     (2, 128, 256, 1024, 64),      # many chunks, chunk < S
     (2, 128, 256, 1024, 100),     # ragged chunks crossing row boundaries
     (1, 257, 256, 1024, 4096),    # single chunk, odd S
 ])
-def test_fused_matches_unfused(B, S, H, V, chunk):
+def test_fused_matches_unfused(B, S, H, V, chunk, mode, monkeypatch):
+    monkeypatch.setenv("DTGA_CE_MODE", mode)
+    _check_fused(B, S, H, V, chunk)
+
+
+def _check_fused(B, S, H, V, chunk):
     from distributed_training_guide_amd.ops.cross_entropy import \
         causal_lm_loss
     from distributed_training_guide_amd.ops.fused_linear_ce import \
