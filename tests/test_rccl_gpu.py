@@ -94,9 +94,17 @@ def test_ddp_forced_collectives_world1(nccl_world1, monkeypatch):
 
 
 def _run_bench(par, extra=()):
+    import sys as _sys
+
+    _sys.path.insert(0, str(REPO / "tests"))
+    from utils_dist import free_port
+
     env = dict(os.environ)
     env["DTGA_FORCE_COLLECTIVES"] = "1"
     env.pop("RANK", None), env.pop("WORLD_SIZE", None)
+    # never inherit the fixture's rendezvous port (EADDRINUSE)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    env["MASTER_PORT"] = str(free_port())
     out = subprocess.run(
         [sys.executable, str(REPO / "bench.py"), "--parallelism", par,
          "--model", "llama-60m", "--batch-size", "2", "--seq-length", "128",
