@@ -36,6 +36,13 @@ void ce_bwd_launch(const void*, const int64_t*, const float*, void*, float,
                    int64_t, int, int, int, int64_t, int64_t, int, hipStream_t);
 void adamw_launch(const void*, int, float, float, float, float, float, float,
                   float, hipStream_t);
+void ln_fwd_launch(const void*, const void*, const void*, void*, float*,
+                   float*, int64_t, int, float, hipStream_t);
+void ln_bwd_launch(const void*, const void*, const void*, const float*,
+                   const float*, void*, float*, float*, void*, void*, int,
+                   int64_t, int, hipStream_t);
+void gelu_fwd_launch(const void*, void*, int64_t, hipStream_t);
+void gelu_bwd_launch(const void*, const void*, void*, int64_t, hipStream_t);
 void attn_fwd_launch(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, int, float, hipStream_t);
 void attn_bwd_launch(const void*, const void*, const void*, const void*,
@@ -228,6 +235,63 @@ torch::Tensor silu_mul_bwd(torch::Tensor dy, torch::Tensor gu) {
   return dgu;
 }
 
+// ---------------- layernorm + gelu (GPT-2 ops) ----------------
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps) {
+  CHECK_BF16_CONTIG(x);
+  CHECK_BF16_CONTIG(w);
+  CHECK_BF16_CONTIG(b);
+  const int H = (int)x.size(-1);
+  const int64_t nrows = x.numel() / H;
+  auto y = torch::empty_like(x);
+  auto opts = x.options().dtype(torch::kFloat);
+  auto mu = torch::empty({nrows}, opts);
+  auto rstd = torch::empty({nrows}, opts);
+  ln_fwd_launch(x.data_ptr(), w.data_ptr(), b.data_ptr(), y.data_ptr(),
+                mu.data_ptr<float>(), rstd.data_ptr<float>(), nrows, H,
+                (float)eps, cur_stream());
+  return {y, mu, rstd};
+}
+
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor mu,
+                                         torch::Tensor rstd) {
+  CHECK_BF16_CONTIG(dy);
+  CHECK_BF16_CONTIG(x);
+  const int H = (int)x.size(-1);
+  const int64_t nrows = x.numel() / H;
+  const int nblocks = (int)std::min<int64_t>(nrows, 2048);
+  auto dx = torch::empty_like(x);
+  auto dw = torch::empty_like(w);
+  auto db = torch::empty_like(w);
+  auto opts = x.options().dtype(torch::kFloat);
+  auto dw_partial = torch::empty({(int64_t)nblocks, (int64_t)H}, opts);
+  auto db_partial = torch::empty({(int64_t)nblocks, (int64_t)H}, opts);
+  ln_bwd_launch(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                mu.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr(),
+                dw_partial.data_ptr<float>(), db_partial.data_ptr<float>(),
+                dw.data_ptr(), db.data_ptr(), nblocks, nrows, H,
+                cur_stream());
+  return {dx, dw, db};
+}
+
+torch::Tensor gelu_fwd(torch::Tensor x) {
+  CHECK_BF16_CONTIG(x);
+  TORCH_CHECK(x.numel() % 8 == 0, "gelu kernel needs numel % 8 == 0");
+  auto y = torch::empty_like(x);
+  gelu_fwd_launch(x.data_ptr(), y.data_ptr(), x.numel(), cur_stream());
+  return y;
+}
+
+torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x) {
+  CHECK_BF16_CONTIG(dy);
+  CHECK_BF16_CONTIG(x);
+  auto dx = torch::empty_like(x);
+  gelu_bwd_launch(dy.data_ptr(), x.data_ptr(), dx.data_ptr(), x.numel(),
+                  cur_stream());
+  return dx;
+}
+
 // ---------------- cross entropy ----------------
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels,
                                   int64_t S_out, int64_t ignore_index) {
@@ -346,6 +410,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("qkv_rope_bwd", &qkv_rope_bwd);
   m.def("silu_mul_fwd", &silu_mul_fwd);
   m.def("silu_mul_bwd", &silu_mul_bwd);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("gelu_fwd", &gelu_fwd);
+  m.def("gelu_bwd", &gelu_bwd);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_fwd_sharded", &ce_fwd_sharded);
   m.def("ce_bwd", &ce_bwd);

@@ -2,9 +2,9 @@
 /root/reference/01-single-gpu/README.md:9-12 trains `gpt2` = 124M).
 
 Learned positional embeddings, pre-LayerNorm blocks, GELU MLP, tied
-embeddings — config-compatible shapes with HF gpt2.  Attention uses the same
-gfx950 flash kernel (head_dim 64); LayerNorm uses torch's op (not a Llama
-hot-path op; the flagship chapters are Llama).
+embeddings — config-compatible shapes with HF gpt2.  Fully on the in-repo
+gfx950 kernels: the flash attention kernel (head_dim 64), fused LayerNorm
+and tanh-GELU (layernorm.hip) and the fused lm_head+CE loss.
 """
 from dataclasses import dataclass, field
 
@@ -13,6 +13,8 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops import causal_lm_loss, flash_attention
+from ..ops.fused_linear_ce import fused_causal_lm_loss
+from ..ops.layernorm import LayerNorm, gelu
 from .llama import CausalLMOutput, pad_vocab
 
 
@@ -45,12 +47,12 @@ class GPT2Block(nn.Module):
     def __init__(self, config: GPT2Config, device=None, dtype=None):
         super().__init__()
         h = config.hidden_size
-        self.ln_1 = nn.LayerNorm(h, eps=config.layer_norm_epsilon,
-                                 device=device, dtype=dtype)
+        self.ln_1 = LayerNorm(h, eps=config.layer_norm_epsilon,
+                              device=device, dtype=dtype)
         self.qkv_proj = nn.Linear(h, 3 * h, device=device, dtype=dtype)
         self.o_proj = nn.Linear(h, h, device=device, dtype=dtype)
-        self.ln_2 = nn.LayerNorm(h, eps=config.layer_norm_epsilon,
-                                 device=device, dtype=dtype)
+        self.ln_2 = LayerNorm(h, eps=config.layer_norm_epsilon,
+                              device=device, dtype=dtype)
         self.fc_in = nn.Linear(h, config.intermediate_size, device=device,
                                dtype=dtype)
         self.fc_out = nn.Linear(config.intermediate_size, h, device=device,
@@ -67,8 +69,7 @@ class GPT2Block(nn.Module):
         v = v.view(B, S, self.n_heads, self.head_dim).contiguous()
         o = flash_attention(q, k, v)
         x = x + self.o_proj(o.reshape(B, S, h))
-        x = x + self.fc_out(F.gelu(self.fc_in(self.ln_2(x)),
-                                   approximate="tanh"))
+        x = x + self.fc_out(gelu(self.fc_in(self.ln_2(x))))
         return x
 
 
@@ -84,8 +85,8 @@ class GPT2ForCausalLM(nn.Module):
         self.blocks = nn.ModuleList(
             GPT2Block(config, device, dtype)
             for _ in range(config.num_hidden_layers))
-        self.ln_f = nn.LayerNorm(h, eps=config.layer_norm_epsilon,
-                                 device=device, dtype=dtype)
+        self.ln_f = LayerNorm(h, eps=config.layer_norm_epsilon,
+                              device=device, dtype=dtype)
         self.lm_head = nn.Linear(h, v, bias=False, device=device, dtype=dtype)
         if config.tie_word_embeddings:
             self.lm_head.weight = self.wte.weight
@@ -101,10 +102,8 @@ class GPT2ForCausalLM(nn.Module):
                     if isinstance(m, nn.Linear) and m.bias is not None \
                             and not m.bias.is_meta:
                         m.bias.zero_()
-                elif isinstance(m, nn.LayerNorm):
-                    if not m.weight.is_meta:
-                        m.weight.fill_(1.0)
-                        m.bias.zero_()
+                elif isinstance(m, LayerNorm):
+                    m.reset_parameters()
 
     def reset_param_by_name(self, name: str, tensor: torch.Tensor):
         std = self.config.initializer_range
@@ -130,6 +129,12 @@ class GPT2ForCausalLM(nn.Module):
         for block in self.blocks:
             x = block(x)
         x = self.ln_f(x)
+        if labels is not None and x.is_cuda:
+            # fused projection+CE (no retained [B,S,V] grad; see
+            # ops/fused_linear_ce.py) — logits are 65x the hidden size
+            # for gpt2, the dominant activation
+            loss = fused_causal_lm_loss(x, self.lm_head.weight, labels)
+            return CausalLMOutput(loss=loss, logits=None)
         logits = self.lm_head(x)
         loss = None
         if labels is not None:

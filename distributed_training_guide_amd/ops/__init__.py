@@ -2,6 +2,8 @@
 from .adamw import FusedAdamW
 from .attention import flash_attention
 from .cross_entropy import causal_lm_loss, sharded_causal_lm_loss
+from .fused_linear_ce import fused_causal_lm_loss
+from .layernorm import LayerNorm, gelu
 from .rmsnorm import RMSNorm, rmsnorm
 from .rope import qkv_rope, rope
 from .swiglu import silu_mul
@@ -11,6 +13,9 @@ __all__ = [
     "flash_attention",
     "causal_lm_loss",
     "sharded_causal_lm_loss",
+    "fused_causal_lm_loss",
+    "LayerNorm",
+    "gelu",
     "RMSNorm",
     "rmsnorm",
     "qkv_rope",
