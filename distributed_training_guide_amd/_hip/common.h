@@ -10,6 +10,28 @@
 
 #define WAVE 64
 
+// Debug-build device assertions (SURVEY.md §5 "race detection /
+// sanitizers": the compute-sanitizer analogue for this kernel set).
+// Enable with DTGA_HIP_FLAGS="-DDTGA_DEBUG" and a rebuild (touch the
+// sources or rm build/); zero cost in release builds.  A failed assert
+// prints its site and traps the wavefront, which surfaces as a HIP
+// error on the next synchronize instead of silent corruption.
+#ifdef DTGA_DEBUG
+#define DTGA_KERNEL_ASSERT(cond)                                         \
+  do {                                                                   \
+    if (!(cond)) {                                                       \
+      printf("DTGA_KERNEL_ASSERT failed: %s at %s:%d (block %d tid %d)\n", \
+             #cond, __FILE__, __LINE__, (int)blockIdx.x,                 \
+             (int)threadIdx.x);                                          \
+      __builtin_trap();                                                  \
+    }                                                                    \
+  } while (0)
+#else
+#define DTGA_KERNEL_ASSERT(cond) \
+  do {                           \
+  } while (0)
+#endif
+
 using bf16 = __hip_bfloat16;
 typedef __attribute__((ext_vector_type(2))) float f32x2;
 typedef __attribute__((ext_vector_type(4))) float f32x4;

@@ -63,6 +63,7 @@ __global__ void __launch_bounds__(256) ce_fwd_kernel(
           loss[r] = 0.0f;
           lse[r] = -INFINITY;  // marks "ignored" for the backward
         } else {
+          DTGA_KERNEL_ASSERT(label >= 0 && label < V);
           float l = gm + __logf(sum);
           lse[r] = l;
           loss[r] = l - bf2f(xr[label]);
@@ -89,6 +90,7 @@ __global__ void __launch_bounds__(256) ce_bwd_kernel(
     int64_t label = labels[b * S_logits + s + (S_logits - S_out)];
     float l = lse[r];
     bool ignored = (label == ignore_index) || (l == -INFINITY);
+    if (!SHARDED) DTGA_KERNEL_ASSERT(ignored || (label >= 0 && label < V));
     int64_t local = label - vocab_start;
     for (int i = threadIdx.x * 8; i < V; i += blockDim.x * 8) {
       if (i + 8 <= V) {

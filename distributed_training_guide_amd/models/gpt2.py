@@ -42,6 +42,14 @@ class GPT2Config:
     def padded_vocab_size(self) -> int:
         return pad_vocab(self.vocab_size)
 
+    def num_parameters(self) -> int:
+        h = self.hidden_size
+        per_block = 12 * h * h + 13 * h  # qkv/o/fc GEMMs+bias, 2 LN
+        embed = self.padded_vocab_size * h * \
+            (1 if self.tie_word_embeddings else 2) + \
+            self.max_position_embeddings * h
+        return per_block * self.num_hidden_layers + embed + 2 * h
+
 
 class GPT2Block(nn.Module):
     def __init__(self, config: GPT2Config, device=None, dtype=None):

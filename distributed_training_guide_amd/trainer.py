@@ -94,13 +94,23 @@ def run_training(args, strategy):
 
     config = get_config(args.model_name)
     model, optimizer, lr_scheduler = strategy.build(config, args)
-    n_params = sum(p.numel() for p in model.parameters())
-    LOGGER.info(f"Training {n_params * getattr(strategy, 'param_factor', 1)} "
-                f"model parameters (local tensors: {n_params})")
+    n_local = sum(p.numel() for p in model.parameters())
+    # under sharding the local tensor count is NOT the model size; the
+    # config knows the true total (reference logs the full count, 01:52)
+    n_total = config.num_parameters() if hasattr(config, "num_parameters") \
+        else n_local
+    LOGGER.info(f"Training {n_total} model parameters "
+                f"(local tensors: {n_local})")
     LOGGER.info(f"Initialized model uses "
                 f"{get_mem_stats(strategy.device)['curr_alloc_gb']:.3f}gb")
 
-    train_data = load_and_preprocess_data(args, config)
+    # local-rank-0 of each node loads first so it populates any node-local
+    # dataset/tokenizer cache before its peers read it (reference 02:72-73
+    # rank0_first; 06:130 rank_ordered for multi-node HF_HOME)
+    from .parallel.pg import rank_ordered
+
+    with rank_ordered(should_go_first=getattr(strategy, "local_rank", 0) == 0):
+        train_data = load_and_preprocess_data(args, config)
     LOGGER.info(f"{len(train_data)} training samples")
 
     sampler = None
@@ -161,6 +171,11 @@ def run_training(args, strategy):
             sampler.set_epoch(state["epoch"])
         batches = iter(dataloader)
         n_batches = len(dataloader)
+        # rank-0 progress bar (reference 02:132)
+        progress_bar = _tqdm(range(n_batches // accum),
+                             disable=strategy.rank > 0,
+                             initial=min(state["epoch_step"],
+                                         n_batches // accum))
 
         for i_step in range(n_batches // accum):
             if measure_waiting:
@@ -204,6 +219,7 @@ def run_training(args, strategy):
             state["global_step"] += 1
             state["epoch_step"] += 1
             state["running_loss"] += total_loss
+            progress_bar.update(1)
 
             if state["global_step"] % args.log_freq == 0:
                 ms_per_step = sum(t.avg_elapsed_ms() for t in timers.values())
@@ -245,6 +261,20 @@ def run_training(args, strategy):
     if wandb_run is not None:
         wandb_run.finish()
     return state
+
+
+def _tqdm(iterable, disable=False, initial=0):
+    try:
+        import tqdm
+
+        return tqdm.tqdm(iterable, disable=disable, initial=initial)
+    except ImportError:  # progress display is never load-bearing
+
+        class _Noop:
+            def update(self, *_):
+                pass
+
+        return _Noop()
 
 
 def _maybe_init_wandb(args, strategy, resumed):
