@@ -54,7 +54,15 @@ class FusedAdamW(torch.optim.Optimizer):
                     for p in params)
         cached = self._desc_cache.get(gi)
         if cached is None or cached[0] != key:
-            rows = []
+            # grad addresses change whenever zero_grad(set_to_none=True)
+            # lets autograd re-allocate, so this rebuild can run EVERY
+            # step — build the [nchunks, 7] descriptor table with
+            # vectorized numpy per param (a Python list-of-lists +
+            # torch.tensor took ~10 ms for an 8B model's ~30k chunks,
+            # a visible GPU idle bubble before the update kernel).
+            import numpy as np
+
+            blocks = []
             for p in params:
                 g = p.grad
                 if not g.is_contiguous():
@@ -67,18 +75,20 @@ class FusedAdamW(torch.optim.Optimizer):
                     raise RuntimeError(f"unsupported param dtype {p.dtype}")
                 esz_p = 2 if p_bf else 4
                 esz_g = 2 if g_bf else 4
-                for off in range(0, n, CHUNK):
-                    cn = min(CHUNK, n - off)
-                    rows.append([
-                        p.data_ptr() + off * esz_p,
-                        g.data_ptr() + off * esz_g,
-                        st["exp_avg"].data_ptr() + off * 4,
-                        st["exp_avg_sq"].data_ptr() + off * 4,
-                        cn, p_bf, g_bf,
-                    ])
-            desc = torch.tensor(rows, dtype=torch.int64).to(
-                params[0].device, non_blocking=True)
-            self._desc_cache[gi] = (key, desc, len(rows))
+                offs = np.arange(0, n, CHUNK, dtype=np.int64)
+                rows = np.empty((len(offs), 7), dtype=np.int64)
+                rows[:, 0] = p.data_ptr() + offs * esz_p
+                rows[:, 1] = g.data_ptr() + offs * esz_g
+                rows[:, 2] = st["exp_avg"].data_ptr() + offs * 4
+                rows[:, 3] = st["exp_avg_sq"].data_ptr() + offs * 4
+                rows[:, 4] = np.minimum(CHUNK, n - offs)
+                rows[:, 5] = p_bf
+                rows[:, 6] = g_bf
+                blocks.append(rows)
+            table = np.concatenate(blocks)
+            desc = torch.from_numpy(table).to(params[0].device,
+                                              non_blocking=True)
+            self._desc_cache[gi] = (key, desc, len(table))
         _, desc, nchunks = self._desc_cache[gi]
         b1, b2 = group["betas"]
         ext().adamw_step(desc, nchunks, group["lr"], b1, b2, group["eps"],
