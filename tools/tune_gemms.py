@@ -43,7 +43,7 @@ def main():
     tunable.enable(True)
     tunable.tuning_enable(True)
     tunable.set_filename(args.out)
-    tunable.set_max_tuning_duration_ms(args.max_ms)
+    tunable.set_max_tuning_duration(args.max_ms)
     # tune worst-first (wgrad family first)
     order = sorted(SHAPES, key=lambda s: base[s[0]][0])
     t0 = time.time()
