@@ -43,6 +43,10 @@ void ln_bwd_launch(const void*, const void*, const void*, const float*,
                    int64_t, int, hipStream_t);
 void gelu_fwd_launch(const void*, void*, int64_t, hipStream_t);
 void gelu_bwd_launch(const void*, const void*, void*, int64_t, hipStream_t);
+void embed_fwd_launch(const void*, const int64_t*, void*, int64_t, int,
+                      int64_t, hipStream_t);
+void embed_bwd_launch(const void*, const int64_t*, float*, void*, int64_t,
+                      int, int64_t, hipStream_t);
 void attn_fwd_launch(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, int, float, hipStream_t);
 void attn_bwd_launch(const void*, const void*, const void*, const void*,
@@ -235,6 +239,34 @@ torch::Tensor silu_mul_bwd(torch::Tensor dy, torch::Tensor gu) {
   return dgu;
 }
 
+// ---------------- embedding ----------------
+torch::Tensor embedding_fwd(torch::Tensor table, torch::Tensor ids) {
+  CHECK_BF16_CONTIG(table);
+  TORCH_CHECK(ids.dtype() == torch::kLong && ids.is_contiguous());
+  const int64_t V = table.size(0);
+  const int H = (int)table.size(1);
+  auto sizes = ids.sizes().vec();
+  sizes.push_back(H);
+  auto out = torch::empty(sizes, table.options());
+  embed_fwd_launch(table.data_ptr(), ids.data_ptr<int64_t>(),
+                   out.data_ptr(), ids.numel(), H, V, cur_stream());
+  return out;
+}
+
+torch::Tensor embedding_bwd(torch::Tensor dy, torch::Tensor ids, int64_t V) {
+  CHECK_BF16_CONTIG(dy);
+  TORCH_CHECK(ids.dtype() == torch::kLong && ids.is_contiguous());
+  const int H = (int)dy.size(-1);
+  TORCH_CHECK(dy.numel() == ids.numel() * H, "dy/ids shape mismatch");
+  auto acc = torch::empty({V, (int64_t)H},
+                          dy.options().dtype(torch::kFloat));
+  auto dtable = torch::empty({V, (int64_t)H}, dy.options());
+  embed_bwd_launch(dy.data_ptr(), ids.data_ptr<int64_t>(),
+                   acc.data_ptr<float>(), dtable.data_ptr(), ids.numel(), H,
+                   V, cur_stream());
+  return dtable;
+}
+
 // ---------------- layernorm + gelu (GPT-2 ops) ----------------
 std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
                                          torch::Tensor b, double eps) {
@@ -410,6 +442,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("qkv_rope_bwd", &qkv_rope_bwd);
   m.def("silu_mul_fwd", &silu_mul_fwd);
   m.def("silu_mul_bwd", &silu_mul_bwd);
+  m.def("embedding_fwd", &embedding_fwd);
+  m.def("embedding_bwd", &embedding_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("gelu_fwd", &gelu_fwd);

@@ -13,6 +13,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops import causal_lm_loss, flash_attention
+from ..ops.embedding import Embedding
 from ..ops.fused_linear_ce import fused_causal_lm_loss
 from ..ops.layernorm import LayerNorm, gelu
 from .llama import CausalLMOutput, pad_vocab
@@ -87,9 +88,9 @@ class GPT2ForCausalLM(nn.Module):
         self.config = config
         v = config.padded_vocab_size
         h = config.hidden_size
-        self.wte = nn.Embedding(v, h, device=device, dtype=dtype)
-        self.wpe = nn.Embedding(config.max_position_embeddings, h,
-                                device=device, dtype=dtype)
+        self.wte = Embedding(v, h, device=device, dtype=dtype)
+        self.wpe = Embedding(config.max_position_embeddings, h,
+                             device=device, dtype=dtype)
         self.blocks = nn.ModuleList(
             GPT2Block(config, device, dtype)
             for _ in range(config.num_hidden_layers))
@@ -104,7 +105,7 @@ class GPT2ForCausalLM(nn.Module):
         std = self.config.initializer_range
         with torch.no_grad():
             for m in self.modules():
-                if isinstance(m, (nn.Linear, nn.Embedding)):
+                if isinstance(m, (nn.Linear, nn.Embedding, Embedding)):
                     if not m.weight.is_meta:
                         m.weight.normal_(0.0, std)
                     if isinstance(m, nn.Linear) and m.bias is not None \

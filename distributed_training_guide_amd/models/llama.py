@@ -22,6 +22,7 @@ import torch.nn as nn
 
 from ..ops import (RMSNorm, causal_lm_loss, flash_attention, qkv_rope,
                    silu_mul)
+from ..ops.embedding import Embedding
 from ..ops.fused_linear_ce import fused_causal_lm_loss
 from ..ops.rmsnorm import add_rmsnorm
 
@@ -149,7 +150,7 @@ class LlamaForCausalLM(nn.Module):
         self.config = config
         v = config.padded_vocab_size
         h = config.hidden_size
-        self.embed_tokens = nn.Embedding(v, h, device=device, dtype=dtype)
+        self.embed_tokens = Embedding(v, h, device=device, dtype=dtype)
         self.layers = nn.ModuleList(
             LlamaDecoderLayer(config, device, dtype)
             for _ in range(config.num_hidden_layers))
@@ -181,7 +182,7 @@ class LlamaForCausalLM(nn.Module):
 
     @staticmethod
     def _reset_module(m, std):
-        if isinstance(m, (nn.Linear, nn.Embedding)):
+        if isinstance(m, (nn.Linear, nn.Embedding, Embedding)):
             if not m.weight.is_meta:
                 m.weight.normal_(0.0, std)
             if isinstance(m, nn.Linear) and m.bias is not None \
