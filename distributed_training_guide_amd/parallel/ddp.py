@@ -143,8 +143,16 @@ class DistributedDataParallel(nn.Module):
             b.pending -= 1
             if b.pending == 0:
                 b.flat.div_(self.world_size)
-                b.work = dist.all_reduce(b.flat, group=self.group,
-                                         async_op=True)
+                from . import xgmi
+
+                if b.flat.is_cuda and xgmi.algo() == "direct":
+                    # fully-connected xGMI RS+AG (parallel/xgmi.py)
+                    b.work = xgmi.direct_all_reduce(b.flat,
+                                                    group=self.group,
+                                                    async_op=True)
+                else:
+                    b.work = dist.all_reduce(b.flat, group=self.group,
+                                             async_op=True)
                 self._works.append(b.work)
         return hook
 

@@ -191,8 +191,14 @@ class _FSDPUnit:
             self.is_unsharded = True
             return
         if self.flat.is_cuda:
-            self._gather_work = dist.all_gather_into_tensor(
-                self.flat, src, group=self.group, async_op=True)
+            from . import xgmi
+
+            if xgmi.algo() == "direct":
+                self._gather_work = xgmi.direct_all_gather_into(
+                    self.flat, src, group=self.group, async_op=True)
+            else:
+                self._gather_work = dist.all_gather_into_tensor(
+                    self.flat, src, group=self.group, async_op=True)
         else:  # gloo
             chunks = list(self.flat.chunk(self.world))
             self._gather_work = dist.all_gather(chunks, src.clone(),
@@ -244,8 +250,14 @@ class _FSDPUnit:
             out.copy_(flat_g)
             self._rs_work, self._rs_out = None, out
         elif flat_g.is_cuda:
-            self._rs_work = dist.reduce_scatter_tensor(
-                out, flat_g, group=self.group, async_op=True)
+            from . import xgmi
+
+            if xgmi.algo() == "direct":
+                self._rs_work = xgmi.direct_reduce_scatter(
+                    out, flat_g, group=self.group, async_op=True)
+            else:
+                self._rs_work = dist.reduce_scatter_tensor(
+                    out, flat_g, group=self.group, async_op=True)
             self._rs_out = out
         else:  # gloo has no reduce_scatter: all-reduce then slice
             self._rs_work = dist.all_reduce(flat_g, group=self.group,

@@ -125,6 +125,21 @@ def bench_one(name, nbytes, dtype, device, world, rank, iters, warmup):
     rec("broadcast",
         _time_op(lambda: dist.broadcast(full, group_src=0), device, iters,
                  warmup), 1.0)
+
+    # direct (fully-connected xGMI) algorithms — parallel/xgmi.py; ring
+    # vs direct is the measured choice DTGA_XGMI_ALGO selects
+    from distributed_training_guide_amd.parallel import xgmi
+
+    rec("direct_all_reduce",
+        _time_op(lambda: xgmi.direct_all_reduce(full), device, iters,
+                 warmup), 2 * (world - 1) / world if world > 1 else 1.0)
+    rec("direct_all_gather",
+        _time_op(lambda: xgmi.direct_all_gather_into(full, shard), device,
+                 iters, warmup), (world - 1) / world if world > 1 else 1.0)
+    rec("direct_reduce_scatter",
+        _time_op(lambda: xgmi.direct_reduce_scatter(out_shard, full),
+                 device, iters, warmup),
+        (world - 1) / world if world > 1 else 1.0)
     del full, shard, out_shard
     return results
 
