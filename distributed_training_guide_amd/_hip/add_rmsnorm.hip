@@ -123,6 +123,81 @@ __global__ void __launch_bounds__(256) add_rmsnorm_bwd_kernel(
   }
 }
 
+// wave-per-row backward (H % 512 == 0, H <= 4096): see rmsnorm.hip —
+// registers stage dy/x between the two passes, w cached across rows,
+// wave-shuffle reduction, no LDS/barriers; dres_out streams in pass 2.
+__device__ __forceinline__ float wave_sum_ar(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    v += __shfl_xor(v, off, WAVE);
+  return v;
+}
+
+__global__ void __launch_bounds__(256) add_rmsnorm_bwd_wave_kernel(
+    const short* __restrict__ dy, const short* __restrict__ dres_out,
+    const short* __restrict__ x, const short* __restrict__ w,
+    const float* __restrict__ rstd, short* __restrict__ dx,
+    float* __restrict__ dw_partial, int64_t nrows, int H) {
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int nch = H / (WAVE * 8);  // <= 8
+  s16x8 wv[8];
+#pragma unroll
+  for (int c = 0; c < 8; ++c)
+    if (c < nch)
+      wv[c] = *reinterpret_cast<const s16x8*>(w + c * WAVE * 8 + lane * 8);
+  float dwacc[8][8];
+#pragma unroll
+  for (int c = 0; c < 8; ++c)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dwacc[c][j] = 0.0f;
+
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wave; row < nrows;
+       row += (int64_t)gridDim.x * 4) {
+    const short* dyr = dy + row * H;
+    const short* xr = x + row * H;
+    const short* drr = dres_out + row * H;
+    short* dxr = dx + row * H;
+    const float rs = rstd[row];
+    s16x8 dv[8], xv[8];
+    float dot = 0.0f;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      if (c >= nch) break;
+      const int i = c * WAVE * 8 + lane * 8;
+      dv[c] = *reinterpret_cast<const s16x8*>(dyr + i);
+      xv[c] = *reinterpret_cast<const s16x8*>(xr + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dot += bf2f(dv[c][j]) * bf2f(wv[c][j]) * bf2f(xv[c][j]);
+    }
+    dot = wave_sum_ar(dot) * rs / (float)H;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      if (c >= nch) break;
+      const int i = c * WAVE * 8 + lane * 8;
+      s16x8 rv = *reinterpret_cast<const s16x8*>(drr + i);
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = bf2f(dv[c][j]) * bf2f(wv[c][j]);
+        float xhat = bf2f(xv[c][j]) * rs;
+        o[j] = f2bf(rs * (g - xhat * dot) + bf2f(rv[j]));
+        dwacc[c][j] += bf2f(dv[c][j]) * xhat;
+      }
+      *reinterpret_cast<s16x8*>(dxr + i) = o;
+    }
+  }
+  float* dwp = dw_partial + ((int64_t)blockIdx.x * 4 + wave) * H;
+#pragma unroll
+  for (int c = 0; c < 8; ++c) {
+    if (c >= nch) break;
+    const int i = c * WAVE * 8 + lane * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dwp[i + j] = dwacc[c][j];
+  }
+}
+
 extern "C" {
 void add_rmsnorm_fwd_launch(const void* res, const void* delta, const void* w,
                             void* res_out, void* y, void* rstd, int64_t nrows,
@@ -133,14 +208,29 @@ void add_rmsnorm_fwd_launch(const void* res, const void* delta, const void* w,
                      (const short*)res, (const short*)delta, (const short*)w,
                      (short*)res_out, (short*)y, (float*)rstd, nrows, H, eps);
 }
+void rmsnorm_dw_reduce_launch(const float*, void*, int, int, hipStream_t);
+
 void add_rmsnorm_bwd_launch(const void* dy, const void* dres_out,
                             const void* x, const void* w, const void* rstd,
-                            void* dx, float* dw_partial, int nblocks,
-                            int64_t nrows, int H, hipStream_t s) {
+                            void* dx, float* dw_partial, void* dw,
+                            int nblocks, int64_t nrows, int H,
+                            hipStream_t s) {
+  if ((H & (64 * 8 - 1)) == 0 && H <= 4096 && nblocks >= 4) {
+    int grid = nblocks / 4;
+    int64_t need = (nrows + 3) / 4;
+    if (need < grid) grid = (int)(need < 1 ? 1 : need);
+    hipLaunchKernelGGL(add_rmsnorm_bwd_wave_kernel, dim3(grid), dim3(256),
+                       0, s, (const short*)dy, (const short*)dres_out,
+                       (const short*)x, (const short*)w, (const float*)rstd,
+                       (short*)dx, dw_partial, nrows, H);
+    rmsnorm_dw_reduce_launch(dw_partial, dw, grid * 4, H, s);
+    return;
+  }
   size_t shmem = 2 * (size_t)H * sizeof(short);
   hipLaunchKernelGGL(add_rmsnorm_bwd_kernel, dim3(nblocks), dim3(256), shmem,
                      s, (const short*)dy, (const short*)dres_out,
                      (const short*)x, (const short*)w, (const float*)rstd,
                      (short*)dx, dw_partial, nrows, H);
+  rmsnorm_dw_reduce_launch(dw_partial, dw, nblocks, H, s);
 }
 }

@@ -19,8 +19,8 @@ void qkv_rope_bwd_launch(const void*, const void*, const void*, void*,
 void add_rmsnorm_fwd_launch(const void*, const void*, const void*, void*,
                             void*, void*, int64_t, int, float, hipStream_t);
 void add_rmsnorm_bwd_launch(const void*, const void*, const void*,
-                            const void*, const void*, void*, float*, int,
-                            int64_t, int, hipStream_t);
+                            const void*, const void*, void*, float*, void*,
+                            int, int64_t, int, hipStream_t);
 void rmsnorm_dw_reduce_launch(const float*, void*, int, int, hipStream_t);
 void rope_launch(const void*, void*, const float*, const float*, const int*,
                  int64_t, int, int, int, int, hipStream_t);
@@ -157,10 +157,8 @@ std::vector<torch::Tensor> add_rmsnorm_bwd(torch::Tensor dy,
                    res_out.options().dtype(torch::kFloat));
   add_rmsnorm_bwd_launch(dy.data_ptr(), dres_out.data_ptr(),
                          res_out.data_ptr(), w.data_ptr(), rstd.data_ptr(),
-                         dx.data_ptr(), dw_partial.data_ptr<float>(), nblocks,
-                         nrows, H, cur_stream());
-  rmsnorm_dw_reduce_launch(dw_partial.data_ptr<float>(), dw.data_ptr(),
-                           nblocks, H, cur_stream());
+                         dx.data_ptr(), dw_partial.data_ptr<float>(),
+                         dw.data_ptr(), nblocks, nrows, H, cur_stream());
   return {dx, dw};
 }
 

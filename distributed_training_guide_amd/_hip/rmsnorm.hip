@@ -145,6 +145,83 @@ __global__ void __launch_bounds__(256) rmsnorm_bwd_kernel(
   }
 }
 
+// ---------------- backward, wave-per-row (H % 8 == 0, H <= 4096) -----
+// One 64-lane wave owns a whole row: dy/x staged in REGISTERS between the
+// dot pass and the dx pass (no LDS round trip), w cached in registers
+// across rows, the row reduction is 6 wave shuffles instead of a
+// block_reduce's LDS+barrier ladder.  The block's 4 waves are fully
+// independent — zero __syncthreads in the loop.  dw register partials
+// spill once per WAVE (dw_partial row = blockIdx*4 + wave).
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    v += __shfl_xor(v, off, WAVE);
+  return v;
+}
+
+__global__ void __launch_bounds__(256) rmsnorm_bwd_wave_kernel(
+    const short* __restrict__ dy, const short* __restrict__ x,
+    const short* __restrict__ w, const float* __restrict__ rstd,
+    short* __restrict__ dx, float* __restrict__ dw_partial,
+    int64_t nrows, int H) {
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int nch = H / (WAVE * 8);  // <= 8
+  s16x8 wv[8];
+#pragma unroll
+  for (int c = 0; c < 8; ++c)
+    if (c < nch)
+      wv[c] = *reinterpret_cast<const s16x8*>(w + c * WAVE * 8 + lane * 8);
+  float dwacc[8][8];
+#pragma unroll
+  for (int c = 0; c < 8; ++c)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dwacc[c][j] = 0.0f;
+
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wave; row < nrows;
+       row += (int64_t)gridDim.x * 4) {
+    const short* dyr = dy + row * H;
+    const short* xr = x + row * H;
+    short* dxr = dx + row * H;
+    const float rs = rstd[row];
+    s16x8 dv[8], xv[8];
+    float dot = 0.0f;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      if (c >= nch) break;
+      const int i = c * WAVE * 8 + lane * 8;
+      dv[c] = *reinterpret_cast<const s16x8*>(dyr + i);
+      xv[c] = *reinterpret_cast<const s16x8*>(xr + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dot += bf2f(dv[c][j]) * bf2f(wv[c][j]) * bf2f(xv[c][j]);
+    }
+    dot = wave_sum(dot) * rs / (float)H;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      if (c >= nch) break;
+      const int i = c * WAVE * 8 + lane * 8;
+      s16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = bf2f(dv[c][j]) * bf2f(wv[c][j]);
+        float xhat = bf2f(xv[c][j]) * rs;
+        o[j] = f2bf(rs * (g - xhat * dot));
+        dwacc[c][j] += bf2f(dv[c][j]) * xhat;
+      }
+      *reinterpret_cast<s16x8*>(dxr + i) = o;
+    }
+  }
+  float* dwp = dw_partial + ((int64_t)blockIdx.x * 4 + wave) * H;
+#pragma unroll
+  for (int c = 0; c < 8; ++c) {
+    if (c >= nch) break;
+    const int i = c * WAVE * 8 + lane * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dwp[i + j] = dwacc[c][j];
+  }
+}
+
 // reduce dw partials: [nblocks, H] f32 -> dw [H] bf16.  Two stages so the
 // read of the (up to 2048 x H) partial buffer parallelizes over enough
 // blocks to reach memory bandwidth (a single H/256-block pass is
@@ -196,6 +273,18 @@ void rmsnorm_bwd_launch(const void* dy, const void* x, const void* w,
                         const void* rstd, void* dx, float* dw_partial,
                         void* dw, int nblocks, int64_t nrows, int H,
                         hipStream_t s) {
+  if ((H & (64 * 8 - 1)) == 0 && H <= 4096 && nblocks >= 4) {
+    // wave-per-row fast path: 4 independent waves per block, partial
+    // rows = 4 * grid (fits the caller's [nblocks, H] allocation)
+    int grid = nblocks / 4;
+    int64_t need = (nrows + 3) / 4;
+    if (need < grid) grid = (int)(need < 1 ? 1 : need);
+    hipLaunchKernelGGL(rmsnorm_bwd_wave_kernel, dim3(grid), dim3(256), 0, s,
+                       (const short*)dy, (const short*)x, (const short*)w,
+                       (const float*)rstd, (short*)dx, dw_partial, nrows, H);
+    rmsnorm_dw_reduce_launch(dw_partial, dw, grid * 4, H, s);
+    return;
+  }
   // scalar fallback (H not a multiple of 8) accumulates into dw_partial
   // directly, so it must start zeroed; the vectorized path overwrites.
   if ((H & 7) || H > 2048 * 8)
