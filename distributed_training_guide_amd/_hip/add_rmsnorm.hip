@@ -12,6 +12,7 @@
 //   dw += dy * xhat   (register partials, one spill per block — same
 //                      scheme as rmsnorm.hip)
 #include "common.h"
+#include <stdlib.h>
 
 __global__ void __launch_bounds__(256) add_rmsnorm_fwd_kernel(
     const short* __restrict__ res, const short* __restrict__ delta,
@@ -215,7 +216,12 @@ void add_rmsnorm_bwd_launch(const void* dy, const void* dres_out,
                             void* dx, float* dw_partial, void* dw,
                             int nblocks, int64_t nrows, int H,
                             hipStream_t s) {
-  if ((H & (64 * 8 - 1)) == 0 && H <= 4096 && nblocks >= 4) {
+  static int wave_ok = -1;
+  if (wave_ok < 0) {  // DTGA_NORM_BLOCK=1 forces the block-per-row path (A/B)
+    const char* e = getenv("DTGA_NORM_BLOCK");
+    wave_ok = !(e && e[0] == '1');
+  }
+  if (wave_ok && (H & (64 * 8 - 1)) == 0 && H <= 4096 && nblocks >= 4) {
     int grid = nblocks / 4;
     int64_t need = (nrows + 3) / 4;
     if (need < grid) grid = (int)(need < 1 ? 1 : need);

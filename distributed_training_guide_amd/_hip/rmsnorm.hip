@@ -9,6 +9,7 @@
 //   dx = rstd * (g - xhat * mean(g * xhat)),  g = dy * w,  xhat = x * rstd
 //   dw = sum_rows dy * xhat   (two-stage: per-block partials, then reduce)
 #include "common.h"
+#include <stdlib.h>
 
 // ---------------- forward ----------------
 __global__ void __launch_bounds__(256) rmsnorm_fwd_kernel(
@@ -273,7 +274,12 @@ void rmsnorm_bwd_launch(const void* dy, const void* x, const void* w,
                         const void* rstd, void* dx, float* dw_partial,
                         void* dw, int nblocks, int64_t nrows, int H,
                         hipStream_t s) {
-  if ((H & (64 * 8 - 1)) == 0 && H <= 4096 && nblocks >= 4) {
+  static int wave_ok = -1;
+  if (wave_ok < 0) {  // DTGA_NORM_BLOCK=1 forces the block-per-row path (A/B)
+    const char* e = getenv("DTGA_NORM_BLOCK");
+    wave_ok = !(e && e[0] == '1');
+  }
+  if (wave_ok && (H & (64 * 8 - 1)) == 0 && H <= 4096 && nblocks >= 4) {
     // wave-per-row fast path: 4 independent waves per block, partial
     // rows = 4 * grid (fits the caller's [nblocks, H] allocation)
     int grid = nblocks / 4;
