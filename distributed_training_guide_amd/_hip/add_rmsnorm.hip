@@ -217,9 +217,11 @@ void add_rmsnorm_bwd_launch(const void* dy, const void* dres_out,
                             int nblocks, int64_t nrows, int H,
                             hipStream_t s) {
   static int wave_ok = -1;
-  if (wave_ok < 0) {  // DTGA_NORM_BLOCK=1 forces the block-per-row path (A/B)
-    const char* e = getenv("DTGA_NORM_BLOCK");
-    wave_ok = !(e && e[0] == '1');
+  if (wave_ok < 0) {  // measured: block-per-row wins at bs24/H=4096 (0.281
+    // vs 0.319 ms — the wave variant's ~200 VGPRs cap occupancy at
+    // 2 waves/SIMD); DTGA_NORM_WAVE=1 selects the wave path for re-eval
+    const char* e = getenv("DTGA_NORM_WAVE");
+    wave_ok = (e && e[0] == '1');
   }
   if (wave_ok && (H & (64 * 8 - 1)) == 0 && H <= 4096 && nblocks >= 4) {
     int grid = nblocks / 4;
