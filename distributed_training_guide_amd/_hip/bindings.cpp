@@ -33,7 +33,8 @@ void ce_fwd_sharded_launch(const void*, const int64_t*, float*, float*, float*,
                            int64_t, int, int, int, int64_t, int64_t,
                            hipStream_t);
 void ce_bwd_launch(const void*, const int64_t*, const float*, void*, float,
-                   int64_t, int, int, int, int64_t, int64_t, int, hipStream_t);
+                   const float*, int64_t, int, int, int, int64_t, int64_t,
+                   int, hipStream_t);
 void adamw_launch(const void*, int, float, float, float, float, float, float,
                   float, hipStream_t);
 void ln_fwd_launch(const void*, const void*, const void*, void*, float*,
@@ -361,16 +362,26 @@ std::vector<torch::Tensor> ce_fwd_sharded(torch::Tensor logits,
 
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels,
                      torch::Tensor lse, double scale, int64_t S_out,
-                     int64_t vocab_start, int64_t ignore_index, bool sharded) {
+                     int64_t vocab_start, int64_t ignore_index, bool sharded,
+                     c10::optional<torch::Tensor> scale_t = c10::nullopt) {
   CHECK_BF16_CONTIG(logits);
   const int B = (int)logits.size(0), S = (int)logits.size(1),
             V = (int)logits.size(2);
   const int64_t nrows = (int64_t)B * S_out;
-  auto dlogits = torch::zeros_like(logits);  // non-loss rows stay zero
+  const float* sp = nullptr;
+  if (scale_t.has_value()) {
+    TORCH_CHECK(scale_t->dtype() == torch::kFloat && scale_t->is_cuda() &&
+                scale_t->numel() == 1, "scale_t must be a cuda f32 scalar");
+    sp = scale_t->data_ptr<float>();
+  }
+  // all rows with S_out == S are loss rows (the kernel writes every one);
+  // otherwise the skipped rows must stay zero
+  auto dlogits = (S_out == S) ? torch::empty_like(logits)
+                              : torch::zeros_like(logits);
   ce_bwd_launch(logits.data_ptr(), labels.data_ptr<int64_t>(),
-                lse.data_ptr<float>(), dlogits.data_ptr(), (float)scale, nrows,
-                S, (int)S_out, V, vocab_start, ignore_index, sharded ? 1 : 0,
-                cur_stream());
+                lse.data_ptr<float>(), dlogits.data_ptr(), (float)scale, sp,
+                nrows, S, (int)S_out, V, vocab_start, ignore_index,
+                sharded ? 1 : 0, cur_stream());
   return dlogits;
 }
 
@@ -448,7 +459,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gelu_bwd", &gelu_bwd);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_fwd_sharded", &ce_fwd_sharded);
-  m.def("ce_bwd", &ce_bwd);
+  m.def("ce_bwd", &ce_bwd, py::arg("logits"), py::arg("labels"),
+        py::arg("lse"), py::arg("scale"), py::arg("S_out"),
+        py::arg("vocab_start"), py::arg("ignore_index"),
+        py::arg("sharded"), py::arg("scale_t") = c10::nullopt);
   m.def("adamw_step", &adamw_step);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);

@@ -80,8 +80,10 @@ template <bool SHARDED>
 __global__ void __launch_bounds__(256) ce_bwd_kernel(
     const short* __restrict__ logits, const int64_t* __restrict__ labels,
     const float* __restrict__ lse, short* __restrict__ dlogits, float scale,
+    const float* __restrict__ scale_ptr,  // device scalar (no host sync)
     int64_t nrows, int S_logits, int S_out, int V,
     int64_t vocab_start, int64_t ignore_index) {
+  if (scale_ptr != nullptr) scale = scale_ptr[0];
   for (int64_t r = blockIdx.x; r < nrows; r += gridDim.x) {
     int64_t b = r / S_out;
     int64_t s = r % S_out;
@@ -137,19 +139,20 @@ void ce_fwd_sharded_launch(const void* logits, const int64_t* labels,
                      ignore_index);
 }
 void ce_bwd_launch(const void* logits, const int64_t* labels, const float* lse,
-                   void* dlogits, float scale, int64_t nrows, int S_logits,
-                   int S_out, int V, int64_t vocab_start, int64_t ignore_index,
-                   int sharded, hipStream_t st) {
+                   void* dlogits, float scale, const float* scale_ptr,
+                   int64_t nrows, int S_logits, int S_out, int V,
+                   int64_t vocab_start, int64_t ignore_index, int sharded,
+                   hipStream_t st) {
   int grid = (int)(nrows < 2048 ? (nrows < 1 ? 1 : nrows) : 2048);
   if (sharded)
     hipLaunchKernelGGL((ce_bwd_kernel<true>), dim3(grid), dim3(256), 0, st,
                        (const short*)logits, labels, lse, (short*)dlogits,
-                       scale, nrows, S_logits, S_out, V, vocab_start,
-                       ignore_index);
+                       scale, scale_ptr, nrows, S_logits, S_out, V,
+                       vocab_start, ignore_index);
   else
     hipLaunchKernelGGL((ce_bwd_kernel<false>), dim3(grid), dim3(256), 0, st,
                        (const short*)logits, labels, lse, (short*)dlogits,
-                       scale, nrows, S_logits, S_out, V, vocab_start,
-                       ignore_index);
+                       scale, scale_ptr, nrows, S_logits, S_out, V,
+                       vocab_start, ignore_index);
 }
 }

@@ -24,19 +24,19 @@ class _CausalCEFn(torch.autograd.Function):
         B, S, V = logits.shape
         S_out = S - 1
         loss_rows, lse = ext().ce_fwd(logits, labels, S_out, IGNORE_INDEX)
-        shifted = labels[:, 1:]
-        n_valid = int((shifted != IGNORE_INDEX).sum().item())
-        ctx.save_for_backward(logits, labels, lse)
-        ctx.n_valid = max(n_valid, 1)
+        # valid-token count stays a DEVICE scalar: no GPU->CPU sync in the
+        # hot loop (and the backward scale is computed on device too)
+        n_valid = (labels[:, 1:] != IGNORE_INDEX).sum().clamp_min(1)
+        ctx.save_for_backward(logits, labels, lse, n_valid)
         ctx.S_out = S_out
-        return loss_rows.sum() / ctx.n_valid
+        return loss_rows.sum() / n_valid
 
     @staticmethod
     def backward(ctx, dloss):
-        logits, labels, lse = ctx.saved_tensors
-        scale = float(dloss.item()) / ctx.n_valid
-        dlogits = ext().ce_bwd(logits, labels, lse, scale, ctx.S_out, 0,
-                               IGNORE_INDEX, False)
+        logits, labels, lse, n_valid = ctx.saved_tensors
+        scale_t = (dloss.detach() / n_valid).to(torch.float32).reshape(1)
+        dlogits = ext().ce_bwd(logits, labels, lse, 0.0, ctx.S_out, 0,
+                               IGNORE_INDEX, False, scale_t.contiguous())
         return dlogits, None
 
 
@@ -66,21 +66,21 @@ class _ShardedCausalCEFn(torch.autograd.Function):
         lse = gmax + torch.log(sm)
         shifted = labels[:, 1:]
         valid = shifted != IGNORE_INDEX
-        n_valid = max(int(valid.sum().item()), 1)
+        n_valid = valid.sum().clamp_min(1)
         loss_rows = torch.where(valid.reshape(-1), lse - gathered,
                                 torch.zeros_like(lse))
-        ctx.save_for_backward(logits, labels, lse)
-        ctx.n_valid = n_valid
+        ctx.save_for_backward(logits, labels, lse, n_valid)
         ctx.S_out = S_out
         ctx.vocab_start = vocab_start
         return loss_rows.sum() / n_valid
 
     @staticmethod
     def backward(ctx, dloss):
-        logits, labels, lse = ctx.saved_tensors
-        scale = float(dloss.item()) / ctx.n_valid
-        dlogits = ext().ce_bwd(logits, labels, lse, scale, ctx.S_out,
-                               ctx.vocab_start, IGNORE_INDEX, True)
+        logits, labels, lse, n_valid = ctx.saved_tensors
+        scale_t = (dloss.detach() / n_valid).to(torch.float32).reshape(1)
+        dlogits = ext().ce_bwd(logits, labels, lse, 0.0, ctx.S_out,
+                               ctx.vocab_start, IGNORE_INDEX, True,
+                               scale_t.contiguous())
         return dlogits, None, None, None
 
 

@@ -52,7 +52,8 @@ class _FusedLinearCEFn(torch.autograd.Function):
         x2 = x.reshape(B * S, H)
         N = B * S
         shifted = _shifted_flat_labels(labels)
-        n_valid = max(int((shifted != IGNORE_INDEX).sum().item()), 1)
+        # device scalar: no GPU->CPU sync per step (see cross_entropy.py)
+        n_valid = (shifted != IGNORE_INDEX).sum().clamp_min(1)
         logits = None
         if keep_logits:
             logits = torch.matmul(x2, weight.t())  # [N,V], saved
@@ -72,10 +73,9 @@ class _FusedLinearCEFn(torch.autograd.Function):
                 loss_sum += lc.sum()
                 lse[r0:r1] = lse_c
         if keep_logits:
-            ctx.save_for_backward(x2, weight, shifted, lse, logits)
+            ctx.save_for_backward(x2, weight, shifted, lse, n_valid, logits)
         else:
-            ctx.save_for_backward(x2, weight, shifted, lse)
-        ctx.n_valid = n_valid
+            ctx.save_for_backward(x2, weight, shifted, lse, n_valid)
         ctx.chunk_rows = chunk_rows
         ctx.in_shape = (B, S, H)
         ctx.keep_logits = keep_logits
@@ -84,14 +84,15 @@ class _FusedLinearCEFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dloss):
         if ctx.keep_logits:
-            x2, weight, shifted, lse, logits = ctx.saved_tensors
+            x2, weight, shifted, lse, n_valid, logits = ctx.saved_tensors
         else:
-            x2, weight, shifted, lse = ctx.saved_tensors
+            x2, weight, shifted, lse, n_valid = ctx.saved_tensors
             logits = None
         B, S, H = ctx.in_shape
         V = weight.shape[0]
         N = B * S
-        scale = float(dloss.item()) / ctx.n_valid
+        scale_t = (dloss.detach() / n_valid).to(torch.float32) \
+            .reshape(1).contiguous()
         dx2 = torch.empty_like(x2)
         one_chunk = ctx.chunk_rows >= N
         # multi-chunk: accumulate dW in fp32 across chunks; single chunk:
@@ -106,8 +107,8 @@ class _FusedLinearCEFn(torch.autograd.Function):
             logits_c = (logits[r0:r1] if logits is not None
                         else torch.matmul(x2[r0:r1], wt))  # recompute
             dlogits_c = ext().ce_bwd(logits_c.unsqueeze(0), shifted[r0:r1],
-                                     lse[r0:r1], scale, L, 0, IGNORE_INDEX,
-                                     False).squeeze(0)
+                                     lse[r0:r1], 0.0, L, 0, IGNORE_INDEX,
+                                     False, scale_t).squeeze(0)
             del logits_c
             torch.matmul(dlogits_c, weight, out=dx2[r0:r1])
             if one_chunk:
