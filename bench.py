@@ -68,6 +68,12 @@ def parse_args():
     p.add_argument("--tensor-parallel", type=int, default=0,
                    help="tp degree for tp/2d (0 = N for tp, 2 for 2d)")
     p.add_argument("--checkpoint-activations", action="store_true")
+    p.add_argument("--hip-graphs", action="store_true",
+                   help="capture the whole training step in one hipGraph "
+                        "and replay it (single-GPU; grads kept allocated; "
+                        "AdamW bias-correction step frozen at capture — "
+                        "a launch-overhead experiment, not the default "
+                        "training path)")
     p.add_argument("--zero1", action="store_true",
                    help="ZeRO-1 optimizer sharding under ddp (off by "
                         "default: on one 288GB-HBM node the replicated "
@@ -232,6 +238,33 @@ def main():
         opt.step()
         opt.zero_grad(set_to_none=True)
         return out.loss
+
+    if args.hip_graphs:
+        if world > 1:
+            raise SystemExit("--hip-graphs is single-GPU only")
+
+        def step_graphable():
+            opt.zero_grad(set_to_none=False)  # stable grad storage
+            out = model(input_ids=ids, labels=ids)
+            out.loss.backward()
+            opt.step()
+            return out.loss
+
+        # warmup on a side stream (torch graph-capture protocol), then
+        # capture one full step; replay is a single hipGraphLaunch
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                step_graphable()
+        torch.cuda.current_stream().wait_stream(side)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            static_loss = step_graphable()
+
+        def step():  # noqa: F811 - graph-replay step
+            graph.replay()
+            return static_loss
 
     def sync():
         if device.type == "cuda":
