@@ -40,8 +40,10 @@ def _shifted_flat_labels(labels: torch.Tensor) -> torch.Tensor:
     flat = labels.reshape(-1)
     out = torch.empty_like(flat)
     out[: B * S - 1] = flat[1:]
-    out[B * S - 1] = IGNORE_INDEX
-    out.view(B, S)[:, S - 1] = IGNORE_INDEX
+    # fill_ (a device kernel) rather than scalar assignment (a pageable
+    # H2D copy): keeps this hipGraph-capturable
+    out[B * S - 1:].fill_(IGNORE_INDEX)
+    out.view(B, S)[:, S - 1].fill_(IGNORE_INDEX)
     return out.contiguous()
 
 
