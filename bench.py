@@ -225,7 +225,9 @@ def main():
 
     _maybe_load_tunableop(model_name, device)
 
-    torch.manual_seed(1234 + rank)
+    # model init seeded IDENTICALLY on every rank (sharded strategies
+    # slice rank-locally from same-seed full tensors); data per-rank
+    torch.manual_seed(1234)
     config = get_config(model_name)
     model, opt, dp_size, label = build(par, args, config, device, world)
 
