@@ -69,6 +69,27 @@ def test_bench_world2(par, label, gbatch):
     assert rec["config"]["global_batch"] == gbatch
 
 
+def test_bench_2d_world4():
+    """dp2 x tp2: both mesh communicator sets live at once (FSDP over the
+    dp dim of TP-sharded layers) — the composition the 8-GPU round-end
+    run exercises at dp4 x tp2."""
+    port = free_port()
+    env = dict(os.environ)
+    env.pop("RANK", None), env.pop("WORLD_SIZE", None)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), str(REPO / "bench.py"),
+         "--gpus", "4", "--parallelism", "2d", "--tensor-parallel", "2"]
+        + COMMON,
+        capture_output=True, text=True, timeout=600, cwd=REPO, env=env)
+    assert out.returncode == 0, out.stderr[-2500:]
+    line = [ln for ln in out.stdout.splitlines() if ln.startswith("{")][-1]
+    rec = _check_contract(line, 4)
+    assert rec["config"]["parallelism"] == "2d_fsdp2_tp2"
+    assert rec["config"]["global_batch"] == 4  # dp=2 replicas x bs2
+
+
 def test_bench_collectives_world2():
     port = free_port()
     env = dict(os.environ)
