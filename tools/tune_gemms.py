@@ -1,9 +1,14 @@
 #!/usr/bin/env python3
 """Offline hipBLASLt/rocBLAS TunableOp search over the training-step GEMM
 shapes (worst first: the TN wgrad family runs 989-1225 TF/s vs 1300-1645
-for fwd — tools/bench_gemm_shapes.py).  Writes the result file that
-bench.py auto-loads (profiles/tunableop_<model>.csv) and prints tuned
-vs untuned per shape.
+for fwd — tools/bench_gemm_shapes.py).  Writes a result file and prints
+tuned vs untuned per shape.
+
+Finding (round 2, profiles/gemm_study_r02.md): the search's internal
+timings do NOT reproduce — loading the results (API or env) leaves every
+shape and the end-to-end step bit-identical to the Default picks.  Kept
+as tooling for future ROCm stacks; the default Tensile selections are
+already at the random-operand wall on this one.
 
     python tools/tune_gemms.py --out gpurun_out/tunableop_llama-3-8b.csv
 """
@@ -50,7 +55,8 @@ def main():
     for name, m, n, k, a_t, b_t in order:
         bench(m, n, k, a_t, b_t, iters=1)
         print(f"tuned {name} ({time.time() - t0:.0f}s elapsed)", flush=True)
-    tunable.write_file()
+    # results are flushed to the filename incrementally / at exit;
+    # torch 2.10 has no tunable.write_file()
     tunable.tuning_enable(False)
 
     print(f"{'shape':14s} {'untuned':>9s} {'tuned':>9s} {'gain':>7s}")
