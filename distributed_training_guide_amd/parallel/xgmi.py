@@ -57,6 +57,8 @@ def direct_all_gather_into(out: torch.Tensor, shard: torch.Tensor,
     exchanges simultaneously (one batched p2p group)."""
     world = dist.get_world_size(group)
     rank = dist.get_rank(group)
+    assert out.numel() == shard.numel() * world, \
+        "direct_all_gather_into: out must be world x shard"
     chunks = list(out.chunk(world))
     if world == 1:
         chunks[0].copy_(shard)
@@ -84,6 +86,8 @@ def direct_reduce_scatter(out: torch.Tensor, inp: torch.Tensor, group=None,
     exchange."""
     world = dist.get_world_size(group)
     rank = dist.get_rank(group)
+    assert inp.numel() == out.numel() * world, \
+        "direct_reduce_scatter: inp must be world x out"
     chunks = list(inp.chunk(world))
     if world == 1:
         out.copy_(chunks[0])
