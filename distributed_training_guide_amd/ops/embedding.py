@@ -42,7 +42,13 @@ class Embedding(nn.Module):
     def forward(self, ids):
         import os
 
-        if use_hip(self.weight) and not os.environ.get("DTGA_TORCH_EMBED"):
+        # the scatter-add backward uses fp32 atomics (reduction order is
+        # run-dependent): under the determinism recipe
+        # (torch.use_deterministic_algorithms, --deterministic) route to
+        # torch's sort-based deterministic embedding backward instead
+        if (use_hip(self.weight)
+                and not os.environ.get("DTGA_TORCH_EMBED")
+                and not torch.are_deterministic_algorithms_enabled()):
             return _EmbeddingFn.apply(ids.contiguous(), self.weight)
         return F.embedding(ids, self.weight)
 
