@@ -62,6 +62,37 @@ def _check_fused(B, S, H, V, chunk):
     assert abs(loss_f.item() - ref.item()) / ref.item() < 2e-2
 
 
+@pytest.mark.parametrize("mode", ["semi", "fused", "unfused"])
+def test_train_steps_under_each_ce_mode(mode, monkeypatch):
+    """Two full train steps (fwd+bwd+FusedAdamW) per CE mode: losses must
+    agree across modes to bf16 tolerance and stay finite."""
+    from distributed_training_guide_amd.models import build_model
+    from distributed_training_guide_amd.ops import FusedAdamW
+
+    monkeypatch.setenv("DTGA_CE_MODE", mode)
+    monkeypatch.setenv("DTGA_CE_CHUNK", "100")  # force ragged chunks
+    torch.manual_seed(11)
+    m = build_model("llama-debug", device=torch.device("cuda"),
+                    dtype=torch.bfloat16)
+    opt = FusedAdamW(m.parameters(), lr=1e-4)
+    ids = torch.randint(0, 1024, (2, 96), device="cuda")
+    losses = []
+    for _ in range(2):
+        out = m(input_ids=ids, labels=ids)
+        losses.append(out.loss.item())
+        out.loss.backward()
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+    torch.cuda.synchronize()
+    assert all(x == x for x in losses)
+    if not hasattr(test_train_steps_under_each_ce_mode, "_ref"):
+        test_train_steps_under_each_ce_mode._ref = losses
+    else:
+        ref = test_train_steps_under_each_ce_mode._ref
+        for a, b in zip(losses, ref):
+            assert abs(a - b) / max(abs(b), 1e-6) < 2e-2, (mode, losses, ref)
+
+
 def test_model_forward_uses_fused_path():
     """llama forward with labels on GPU returns logits=None (the fused
     path) and a finite loss that matches the CPU fp32 full path."""
