@@ -9,7 +9,6 @@ single-rank: gathers are device copies on the comm stream) or under
 torchrun at world>1 for real ring kernels.
 """
 import os
-import sys
 import time
 
 import torch
