@@ -114,3 +114,20 @@ def test_add_rmsnorm_cpu_fallback():
     y, res_out = add_rmsnorm(res, delta, w, 1e-5)
     assert torch.allclose(res_out, res + delta)
     assert torch.allclose(y, rmsnorm_ref(res + delta, w, 1e-5), atol=1e-5)
+
+
+def test_shifted_flat_labels():
+    """Causal shift folded into a flat label vector: out[b*S+s] =
+    labels[b,s+1], last position of each row = ignore_index."""
+    import torch
+
+    from distributed_training_guide_amd.ops.cross_entropy import IGNORE_INDEX
+    from distributed_training_guide_amd.ops.fused_linear_ce import \
+        _shifted_flat_labels
+
+    labels = torch.arange(12).view(3, 4)
+    out = _shifted_flat_labels(labels).view(3, 4)
+    for b in range(3):
+        for s in range(3):
+            assert out[b, s] == labels[b, s + 1]
+        assert out[b, 3] == IGNORE_INDEX
