@@ -101,3 +101,19 @@ def test_synthetic_determinism():
     assert torch.equal(d1[5]["input_ids"], d2[5]["input_ids"])
     d3 = SyntheticTextDataset(1000, 16, num_samples=8, seed=4)
     assert not torch.equal(d1[5]["input_ids"], d3[5]["input_ids"])
+
+
+def test_packed_token_dataset():
+    """The HF-path packing container: fixed-length int64 rows over one
+    flat stream, trainer row shape (data/pipeline.py)."""
+    from distributed_training_guide_amd.data.pipeline import \
+        PackedTokenDataset
+
+    stream = torch.arange(10, dtype=torch.int32)
+    ds = PackedTokenDataset(stream, seq_length=4)
+    assert len(ds) == 2  # 10 // 4, tail dropped
+    row = ds[1]
+    assert row["input_ids"].tolist() == [4, 5, 6, 7]
+    assert row["input_ids"].dtype == torch.long
+    assert torch.equal(row["labels"], row["input_ids"])
+    assert row["attention_mask"].sum() == 4
