@@ -116,8 +116,11 @@ def direct_reduce_scatter(out: torch.Tensor, inp: torch.Tensor, group=None,
 
 def direct_all_reduce(t: torch.Tensor, group=None, async_op: bool = False):
     """In-place sum all-reduce as direct RS + direct AG (pads to a
-    world-divisible chunking internally).  The scatter phase starts
-    immediately; the gather phase is issued from wait()."""
+    world-divisible chunking internally).  On CUDA, Work.wait() is only a
+    stream dependency, so the WHOLE RS+reduce+AG pipeline is enqueued
+    eagerly here and overlaps whatever the caller does next (the DDP
+    engine's remaining backward); on gloo wait() host-blocks, so the AG
+    phase is deferred into the returned handle's wait()."""
     world = dist.get_world_size(group)
     if world == 1:
         return _DirectWork([]) if async_op else None
@@ -134,11 +137,20 @@ def direct_all_reduce(t: torch.Tensor, group=None, async_op: bool = False):
                         device=t.device)
     rs = direct_reduce_scatter(shard, buf, group, async_op=True)
 
-    def finish():
-        rs.wait()
-        direct_all_gather_into(buf, shard, group)
-        if pad:
-            flat.copy_(buf[:n])
+    if t.is_cuda:
+        rs.wait()  # stream-ordered only: local adds enqueue now
+        ag = direct_all_gather_into(buf, shard, group, async_op=True)
+
+        def finish():
+            ag.wait()
+            if pad:
+                flat.copy_(buf[:n])
+    else:
+        def finish():
+            rs.wait()
+            direct_all_gather_into(buf, shard, group)
+            if pad:
+                flat.copy_(buf[:n])
 
     w = _DirectWork([], finish)
     if async_op:
