@@ -9,10 +9,19 @@ exchanged with ONE flat all-gather over xGMI (params are packed into equal
 padded shards — a single large collective instead of per-tensor broadcasts,
 which is what the 7-link point-to-point fabric wants).
 """
+import os
+
 import torch
 import torch.distributed as dist
 
 from ..ops import FusedAdamW
+
+
+def _force_collectives():
+    """DTGA_FORCE_COLLECTIVES=1: run the post-step shard all-gather even at
+    world=1 (see parallel/fsdp.py) so 1-GPU runs execute the exact
+    multi-GPU call pattern (in-place all_gather_into_tensor)."""
+    return os.environ.get("DTGA_FORCE_COLLECTIVES") == "1"
 
 
 class ZeroRedundancyOptimizer(torch.optim.Optimizer):
@@ -43,7 +52,7 @@ class ZeroRedundancyOptimizer(torch.optim.Optimizer):
         self.state = self.local_opt.state
         self._all_params = params
 
-        if self.world_size > 1:
+        if self.world_size > 1 or _force_collectives():
             self.shard_numel = max(loads)
             dev = params[0].device
             self.dtype = params[0].dtype
@@ -56,7 +65,9 @@ class ZeroRedundancyOptimizer(torch.optim.Optimizer):
     @torch.no_grad()
     def step(self, closure=None):
         loss = self.local_opt.step(closure)
-        if self.world_size > 1:
+        if self.world_size > 1 or (_force_collectives()
+                                   and getattr(self, "_gather_buf",
+                                               None) is not None):
             my = self._gather_buf[self.rank * self.shard_numel:
                                   (self.rank + 1) * self.shard_numel]
             off = 0

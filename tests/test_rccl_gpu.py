@@ -74,15 +74,17 @@ def test_ddp_forced_collectives_world1(nccl_world1, monkeypatch):
     async bucket all-reduce + engine-callback wait) at world=1."""
     monkeypatch.setenv("DTGA_FORCE_COLLECTIVES", "1")
     from distributed_training_guide_amd.models import build_model
-    from distributed_training_guide_amd.ops import FusedAdamW
     from distributed_training_guide_amd.parallel.ddp import \
         DistributedDataParallel
+    from distributed_training_guide_amd.parallel.zero1 import \
+        ZeroRedundancyOptimizer
 
     torch.manual_seed(0)
     model = build_model("llama-debug", device=torch.device("cuda"),
                         dtype=torch.bfloat16)
     model = DistributedDataParallel(model, bucket_cap_mb=8)
-    opt = FusedAdamW(model.parameters(), lr=1e-4)
+    # ZeRO-1's in-place shard all-gather also runs under the forced flag
+    opt = ZeroRedundancyOptimizer(model.parameters(), lr=1e-4)
     ids = torch.randint(0, 1024, (2, 64), device="cuda")
     for _ in range(2):
         out = model(input_ids=ids, labels=ids)
