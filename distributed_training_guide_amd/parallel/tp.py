@@ -405,3 +405,60 @@ class TPLlamaForCausalLM(nn.Module):
 
     def load_tp_state_dict(self, sd):
         self.load_state_dict({k: v for k, v in sd.items()})
+
+    def reconstruct_full_tensor(self, name: str, parts: list):
+        """Reassemble the FULL tensor for parameter `name` from all
+        tp_old ranks' shards (parts[r] = rank r's tensor), per that
+        module's sharding rule — the reshard-on-load primitive (the
+        reference leans on torch DCP's planner for tp-size changes;
+        SURVEY.md §7 'DCP-compatible sharded checkpoint with resharding')."""
+        mod = dict(self.named_modules())[name.rsplit(".", 1)[0]]
+        tp_old = len(parts)
+        if isinstance(mod, ColwiseLinear):
+            # old shard = concat over segments of that segment's slice
+            full_segs = []
+            off = 0
+            for seg in mod.segments:
+                loc = seg // tp_old
+                full_segs.append(torch.cat([p[off: off + loc]
+                                            for p in parts]))
+                off += loc
+            return torch.cat(full_segs)
+        if isinstance(mod, RowwiseLinear):
+            return torch.cat(parts, dim=1)
+        if isinstance(mod, VocabParallelEmbedding):
+            return torch.cat(parts, dim=0)
+        return parts[0]  # replicated (norm weights)
+
+    def shard_tensor(self, name: str, full: torch.Tensor) -> torch.Tensor:
+        """Slice a FULL tensor down to this rank's shard per parameter
+        `name`'s sharding rule (the inverse of reconstruct_full_tensor);
+        also used to reshard optimizer moments, which follow their
+        parameter's layout."""
+        mod = dict(self.named_modules())[name.rsplit(".", 1)[0]]
+        tr = self.mesh.tp_rank
+        if isinstance(mod, ColwiseLinear):
+            return _shard_rows(full, mod.segments, mod.tp, tr).contiguous()
+        if isinstance(mod, RowwiseLinear):
+            return full.narrow(1, tr * mod.in_local,
+                               mod.in_local).contiguous()
+        if isinstance(mod, VocabParallelEmbedding):
+            return full.narrow(0, tr * mod.vlocal, mod.vlocal).contiguous()
+        return full
+
+    def load_tp_state_dict_resharded(self, shards: list):
+        """Load from a DIFFERENT tp size: `shards` is every old rank's
+        tp_state_dict() in rank order; each full tensor is reconstructed
+        and re-sliced for this model's (tp, tp_rank)."""
+        tr = self.mesh.tp_rank
+        mods = dict(self.named_modules())
+        with torch.no_grad():
+            for name, p in self.named_parameters():
+                full = self.reconstruct_full_tensor(
+                    name, [sd[name] for sd in shards])
+                mod = mods[name.rsplit(".", 1)[0]]
+                if isinstance(mod, (ColwiseLinear, RowwiseLinear,
+                                    VocabParallelEmbedding)):
+                    mod.load_full_weight(full.to(p.device), tr)
+                else:
+                    p.copy_(full.to(p.device, p.dtype))

@@ -62,15 +62,63 @@ class TPStrategy:
                           state, self.rank, self.world_size)
 
     def load_checkpoint(self, exp_dir: Path, model, optimizer, lr_scheduler):
-        msd, osd, state = ckpt.load_sharded(exp_dir, self.rank,
-                                            self.world_size)
         inner = model.module if hasattr(model, "module") else model
-        inner.load_tp_state_dict(msd)
-        optimizer.load_state_dict(osd)
+        try:
+            msd, osd, state = ckpt.load_sharded(exp_dir, self.rank,
+                                                self.world_size)
+            inner.load_tp_state_dict(msd)
+            optimizer.load_state_dict(osd)
+        except RuntimeError as e:
+            if "reshard" not in str(e):
+                raise
+            state = _load_tp_resharding(exp_dir, inner, optimizer)
         sched_sd = torch.load(exp_dir / "lr_scheduler.pt",
                               map_location="cpu", weights_only=True)
         lr_scheduler.load_state_dict(sched_sd)
         return state
+
+
+def _load_tp_resharding(exp_dir: Path, inner, optimizer):
+    """tp-size-changed load (ch6: tp == world): reconstruct every full
+    tensor from the OLD ranks' shard files and re-slice for the new tp —
+    model weights via the module sharding rules, optimizer moments via
+    the same rules (they follow their parameter's layout).  The reference
+    leans on torch DCP's planner for this."""
+    import json
+
+    ckpt_dir = exp_dir / "checkpoint"
+    with open(ckpt_dir / "metadata.json") as fp:
+        meta = json.load(fp)
+    w_old = meta["world_size"]
+    LOGGER.info(f"resharding TP checkpoint from tp={w_old} to "
+                f"{inner.mesh.tp_size}")
+    blobs = [torch.load(ckpt_dir / f"shard_rank{r}.pt", map_location="cpu",
+                        weights_only=True) for r in range(w_old)]
+    inner.load_tp_state_dict_resharded([b["model"] for b in blobs])
+    # optimizer moments: state keyed by param index in parameters() order
+    # (== named_parameters order)
+    names = [n for n, _ in inner.named_parameters()]
+    opt_sds = [b["optimizer"] for b in blobs]
+
+    def get(sd, i):
+        st = sd["state"]
+        return st[i] if i in st else st.get(str(i))
+
+    new_state = {}
+    for i, name in enumerate(names):
+        if get(opt_sds[0], i) is None:
+            continue
+        merged = {}
+        for key in ("exp_avg", "exp_avg_sq"):
+            full = inner.reconstruct_full_tensor(
+                name, [get(sd, i)[key] for sd in opt_sds])
+            merged[key] = inner.shard_tensor(name, full)
+        new_state[i] = merged
+    new_sd = {"state": new_state,
+              "param_groups": optimizer.state_dict()["param_groups"]}
+    optimizer.load_state_dict(new_sd)
+    with open(exp_dir / "state.json") as fp:
+        return json.load(fp)
 
 
 class TwoDStrategy(TPStrategy):

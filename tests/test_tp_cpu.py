@@ -206,3 +206,78 @@ def _tp_dp_grads(rank, world):
 
 def test_tp_dp_2d_grads():
     run_dist(_tp_dp_grads, world_size=4)
+
+
+def _tp2_save(rank, world, tmpdir):
+    from pathlib import Path
+
+    import torch.optim as optim
+
+    from distributed_training_guide_amd.models import get_config
+    from distributed_training_guide_amd.ops import FusedAdamW
+    from distributed_training_guide_amd.parallel.mesh import DeviceMesh2D
+    from distributed_training_guide_amd.parallel.tp import TPLlamaForCausalLM
+    from distributed_training_guide_amd.utils import checkpoint as ckpt
+
+    torch.manual_seed(5)
+    mesh = DeviceMesh2D(tp_size=2)
+    m = TPLlamaForCausalLM(get_config("llama-debug"), mesh)
+    opt = FusedAdamW(m.parameters(), lr=1e-3)
+    g = torch.Generator().manual_seed(3)
+    ids = torch.randint(0, 1024, (2, 32), generator=g)
+    out = m(input_ids=ids, labels=ids)
+    out.loss.backward()
+    opt.step()
+    sched = optim.lr_scheduler.CosineAnnealingLR(opt, T_max=10)
+    ckpt.save_sharded(Path(tmpdir), m.tp_state_dict(),
+                      ckpt.optim_sd_cpu(opt), sched,
+                      {"global_step": 1}, rank, world)
+
+
+def _tp1_load_resharded(rank, world, tmpdir):
+    from pathlib import Path
+
+    from distributed_training_guide_amd.models import get_config
+    from distributed_training_guide_amd.ops import FusedAdamW
+    from distributed_training_guide_amd.parallel.mesh import DeviceMesh2D
+    from distributed_training_guide_amd.parallel.tp import TPLlamaForCausalLM
+    from distributed_training_guide_amd.parallel.tp_strategy import \
+        _load_tp_resharding
+
+    torch.manual_seed(99)  # different init: the load must overwrite it
+    mesh = DeviceMesh2D(tp_size=1)
+    m = TPLlamaForCausalLM(get_config("llama-debug"), mesh)
+    opt = FusedAdamW(m.parameters(), lr=1e-3)
+    state = _load_tp_resharding(Path(tmpdir), m, opt)
+    assert state["global_step"] == 1
+
+    # model weights == tp2-save-time weights == single-model weights of
+    # the SAME post-step values: verify against a manual reconstruction
+    blobs = [torch.load(Path(tmpdir) / "checkpoint" / f"shard_rank{r}.pt",
+                        weights_only=True) for r in range(2)]
+    name = "layers.0.self_attn.qkv_proj.weight"
+    expect = m.reconstruct_full_tensor(
+        name, [b["model"][name] for b in blobs])
+    got = dict(m.named_parameters())[name]
+    assert torch.equal(got.detach(), expect)  # tp=1: full == shard
+    # moments resharded alongside (same layout rule)
+    names = [n for n, _ in m.named_parameters()]
+    i = names.index(name)
+    exp_avg = None
+    for p, stt in opt.state.items():
+        if p is dict(m.named_parameters())[name]:
+            exp_avg = stt["exp_avg"]
+    assert exp_avg is not None
+    m_expect = m.reconstruct_full_tensor(
+        name, [b["optimizer"]["state"][i]["exp_avg"]
+               if i in b["optimizer"]["state"]
+               else b["optimizer"]["state"][str(i)]["exp_avg"]
+               for b in blobs])
+    assert torch.allclose(exp_avg, m_expect)
+
+
+def test_tp_reshard_on_load(tmp_path):
+    """Save at tp=2, load at tp=1 through the resharding path: weights and
+    optimizer moments reconstructed per the module sharding rules."""
+    run_dist(_tp2_save, world_size=2, args=(str(tmp_path),))
+    run_dist(_tp1_load_resharded, world_size=1, args=(str(tmp_path),))
