@@ -122,7 +122,13 @@ def _load_tp_resharding(exp_dir: Path, inner, optimizer):
 
 
 class TwoDStrategy(TPStrategy):
-    """Chapter 7: FSDP over the dp mesh dim layered over TP (07:121-123)."""
+    """Chapter 7: FSDP over the dp mesh dim layered over TP (07:121-123).
+
+    Checkpoints are shard-of-shard (each (dp, tp) rank's FSDP shard of its
+    local tp shard); loading requires the SAME (dp, tp) — 1-D resharding
+    exists for pure FSDP (fsdp_strategy._load_resharding) and pure TP
+    (_load_tp_resharding above), 2-D re-meshing is not implemented (the
+    reference gets it from torch DCP's planner)."""
 
     def __init__(self, args):
         super().__init__(args, tp_size=getattr(args, "tensor_parallel", 8))
