@@ -18,9 +18,20 @@ _LLAMA_CONFIGS = {
     "llama-60m": (512, 1376, 8, 8, 8, 32000, 1e4, 2048),
 }
 
+_GPT2_CONFIGS = {
+    # name -> (hidden, layers, heads); head_dim is 64 throughout
+    "gpt2": (768, 12, 12),
+    "gpt2-medium": (1024, 24, 16),
+    "gpt2-large": (1280, 36, 20),
+    "gpt2-xl": (1600, 48, 25),
+}
+
 _ALIASES = {
     "gpt2": "gpt2",
     "openai-community/gpt2": "gpt2",
+    "openai-community/gpt2-medium": "gpt2-medium",
+    "openai-community/gpt2-large": "gpt2-large",
+    "openai-community/gpt2-xl": "gpt2-xl",
     "meta-llama/llama-2-7b-hf": "llama-2-7b",
     "meta-llama/llama-2-13b-hf": "llama-2-13b",
     "meta-llama/meta-llama-3-8b": "llama-3-8b",
@@ -40,8 +51,10 @@ def resolve_name(name: str) -> str:
 
 def get_config(name: str):
     key = resolve_name(name)
-    if key == "gpt2":
-        return GPT2Config()
+    if key in _GPT2_CONFIGS:
+        h, L, heads = _GPT2_CONFIGS[key]
+        return GPT2Config(hidden_size=h, num_hidden_layers=L,
+                          num_attention_heads=heads)
     if key in _LLAMA_CONFIGS:
         h, i, L, hq, hkv, v, theta, mp = _LLAMA_CONFIGS[key]
         return LlamaConfig(
@@ -51,7 +64,8 @@ def get_config(name: str):
             rope_theta=theta,
         )
     raise ValueError(
-        f"unknown model {name!r}; known: gpt2, {', '.join(_LLAMA_CONFIGS)}")
+        f"unknown model {name!r}; known: "
+        f"{', '.join(_GPT2_CONFIGS)}, {', '.join(_LLAMA_CONFIGS)}")
 
 
 def build_model(name_or_config, device=None, dtype=None):
