@@ -13,6 +13,8 @@ _LLAMA_CONFIGS = {
     "llama-3-8b": (4096, 14336, 32, 32, 8, 128256, 5e5, 8192),
     "llama-3-70b": (8192, 28672, 80, 64, 8, 128256, 5e5, 8192),
     "llama-3-405b": (16384, 53248, 126, 128, 8, 128256, 5e5, 8192),
+    "llama-3.2-1b": (2048, 8192, 16, 32, 8, 128256, 5e5, 8192),
+    "llama-3.2-3b": (3072, 8192, 28, 24, 8, 128256, 5e5, 8192),
     # small configs for tests / smoke runs
     "llama-debug": (256, 688, 4, 4, 2, 1024, 1e4, 2048),
     "llama-60m": (512, 1376, 8, 8, 8, 32000, 1e4, 2048),
@@ -41,6 +43,8 @@ _ALIASES = {
     "meta-llama/llama-3.1-70b": "llama-3-70b",
     "meta-llama/meta-llama-3.1-405b": "llama-3-405b",
     "meta-llama/llama-3.1-405b": "llama-3-405b",
+    "meta-llama/llama-3.2-1b": "llama-3.2-1b",
+    "meta-llama/llama-3.2-3b": "llama-3.2-3b",
 }
 
 
@@ -62,6 +66,7 @@ def get_config(name: str):
             num_hidden_layers=L, num_attention_heads=hq,
             num_key_value_heads=hkv, max_position_embeddings=mp,
             rope_theta=theta,
+            tie_word_embeddings=key.startswith("llama-3.2"),
         )
     raise ValueError(
         f"unknown model {name!r}; known: "
