@@ -18,15 +18,19 @@ def test_toy_elastic_restart(tmp_path):
     env["TOY_STATE_FILE"] = str(tmp_path / "toy-state.json")
     env["TOY_FAIL_AT"] = "3"  # deterministic: rank 0 dies once at iter 3
     # single attempt: the toy's interruptible_barrier keeps the surviving
-    # rank signal-responsive, so teardown+restart is fast and this no
-    # longer needs flake-absorbing retries
+    # rank signal-responsive, so OUR teardown is fast.  A separate,
+    # torch-internal race remains: on restart, gloo connectFullMesh can
+    # read the dead incarnation's address from the store and burn ~100 s
+    # before failing, which torchelastic treats as one more worker
+    # failure — the generous --max-restarts budget lets the run ride
+    # through it (each clean run needs exactly 1 restart).
     proc = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run",
          "--standalone", "--local-addr", "127.0.0.1",
-         "--nproc-per-node", "2", "--max-restarts", "3",
+         "--nproc-per-node", "2", "--max-restarts", "6",
          str(REPO / "related-topics" / "elastic-training" / "toy.py")],
         env=env, cwd=str(tmp_path), capture_output=True, text=True,
-        timeout=180)
+        timeout=400)
     out = proc.stdout + proc.stderr
     assert proc.returncode == 0, out[-2000:]
     assert "done: {'iteration': 20}" in out, out[-2000:]
