@@ -276,8 +276,15 @@ def main():
         if world > 1:
             dist.barrier()
 
-    for _ in range(args.warmup):
-        step()
+    try:
+        for _ in range(args.warmup):
+            step()
+    except torch.OutOfMemoryError:
+        print(f"OOM during warmup at batch-size {B} (model {model_name}, "
+              f"parallelism {par}): try a smaller --batch-size, "
+              "--checkpoint-activations, or DTGA_CE_MODE=fused",
+              file=sys.stderr)
+        raise
     sync()          # drain this rank's device work before aligning hosts
     barrier()
     sync()          # RCCL barrier enqueues device work; drain it too
