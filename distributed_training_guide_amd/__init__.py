@@ -17,4 +17,4 @@ Layout:
     trainer.py shared trainer loop behind every chapter entrypoint
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
