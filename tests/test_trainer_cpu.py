@@ -117,3 +117,21 @@ def test_packed_token_dataset():
     assert row["input_ids"].dtype == torch.long
     assert torch.equal(row["labels"], row["input_ids"])
     assert row["attention_mask"].sum() == 4
+
+
+def test_log_dict_schema(tmp_path, capfd):
+    """The per-log-freq info dict carries the reference's keys
+    (01:155-174): step/lr/loss/epoch bookkeeping, the five memory stats,
+    tokens_per_s and per-phase timers.  (captured from stderr: the
+    trainer's setup_logging(force=True) replaces caplog's handler)"""
+    _run(tmp_path, ["--max-steps", "2", "--log-freq", "2"])
+    err = capfd.readouterr().err
+    dicts = [ln for ln in err.splitlines() if "'global_step'" in ln]
+    assert dicts, "no info dict logged"
+    info = eval(dicts[-1].split("INFO:", 1)[1])  # noqa: S307 - own dict
+    for key in ("global_step", "lr", "running_loss", "epoch",
+                "epoch_progress", "num_batches_remaining", "total_gb",
+                "curr_alloc_gb", "peak_alloc_gb", "curr_resv_gb",
+                "peak_resv_gb", "tokens_per_s", "time/total", "time/data",
+                "time/forward", "time/backward", "time/update"):
+        assert key in info, key
