@@ -40,13 +40,29 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", 1))
     # store-barrier before gloo's address exchange: across torchelastic
     # restarts, out-of-phase re-inits can read a dead incarnation's
-    # listener address and burn the full connect timeout per round
-    # (measured ~15% of restart runs without this; 0/24 with it)
+    # listener address (lowers the race rate but does not eliminate it)
     os.environ.setdefault("TORCH_DIST_INIT_BARRIER", "1")
-    # bounded collectives: when a peer dies, the survivors' barrier must
-    # throw promptly so torchelastic can tear down and restart the group
-    dist.init_process_group("gloo",
-                            timeout=datetime.timedelta(seconds=20))
+    # bounded collectives AND a bounded gloo connect: the connectFullMesh
+    # deadline is ~4x this timeout, so 10 s keeps a raced re-init cheap.
+    # When the race fires anyway (stale peer address -> connection
+    # refused), retry the init IN-PROCESS: each attempt uses a fresh
+    # store prefix, so both workers re-pair without burning a
+    # torchelastic restart (restarting the processes re-enters the same
+    # race; retrying the group init resolves it).
+    for attempt in range(4):
+        try:
+            dist.init_process_group(
+                "gloo", timeout=datetime.timedelta(seconds=10))
+            break
+        except RuntimeError as e:
+            if "connectFullMesh" not in str(e) or attempt == 3:
+                raise
+            try:
+                dist.destroy_process_group()
+            except (RuntimeError, ValueError):
+                pass
+            print(f"[rank={rank}] init race (attempt {attempt}); retrying")
+            time.sleep(1.0 + 0.5 * rank)
     # NOTE: rank/world_size are NOT stable across restarts — reload shared
     # progress from the state file, never from process memory.
     state = {"iteration": 0}
