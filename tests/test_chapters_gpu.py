@@ -65,3 +65,17 @@ def test_chapter4_fsdp_gpu_world1_forced(tmp_path):
     assert (exp / "checkpoint" / "shard_rank0.pt").exists()
     assert (exp / "checkpoint" / "metadata.json").exists()
     assert json.loads((exp / "state.json").read_text())["global_step"] == 2
+
+
+def test_bench_hip_graphs_mode(tmp_path):
+    """--hip-graphs captures the whole step and replays it (opt-in
+    experiment); must produce the normal JSON contract."""
+    out = subprocess.run(
+        [sys.executable, str(REPO / "bench.py"), "--model", "llama-60m",
+         "--batch-size", "4", "--seq-length", "256", "--steps", "3",
+         "--warmup", "1", "--hip-graphs"],
+        capture_output=True, text=True, timeout=300, cwd=REPO, env=_env())
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [ln for ln in out.stdout.splitlines() if ln.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["value"] > 0
